@@ -1,0 +1,181 @@
+"""Flagship benchmark: Parquet→DDP ResNet-18 bf16 streaming training.
+
+This is the BASELINE.json headline config ("Parquet→DataLoader ResNet-18
+DDP bf16 on 8×MI355X, Petastorm-equivalent streaming path"): synthetic
+224×224 uint8 images stored in parquet, streamed by the native row-group
+reader (pyarrow decode threads → pinned ring → side-stream H2D), fused
+HIP normalize to bf16 channels-last, ResNet-18 (random init) fwd+bwd+Adam
+under bf16 autocast, gradients all-reduced by DDP over RCCL/xGMI.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+(N>1 is launched by torch.distributed.run, one rank per GPU; rank/env
+from RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*.)
+
+Rank 0 prints ONE JSON line: whole-job samples/sec over exactly K timed
+steps bracketed by barrier+synchronize on both sides, MAX time over ranks.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+from mi355x_scale.parallel.comm import init_distributed, barrier, destroy
+from mi355x_scale.train import ImageClassifier
+from mi355x_scale.train.datamodule import ImageStreamDataModule
+from mi355x_scale.train.trainer import _TrainStepShim
+from torch.nn.parallel import DistributedDataParallel as DDP
+
+DATA_DIR = os.environ.get("MI355X_BENCH_DATA", "/tmp/mi355x_bench_data")
+
+
+def prepare_data(rows: int, image_hw, rank: int, world: int) -> str:
+    from mi355x_scale.data.generator import write_image_parquet
+    marker = os.path.join(DATA_DIR, ".complete")
+    tag = f"{rows}x{image_hw[0]}"
+    if rank == 0:
+        ok = False
+        if os.path.exists(marker):
+            with open(marker) as f:
+                ok = f.read().strip() == tag
+        if not ok:
+            import shutil
+            shutil.rmtree(DATA_DIR, ignore_errors=True)
+            write_image_parquet(DATA_DIR, num_rows=rows, image_hw=image_hw,
+                                rows_per_group=64, rows_per_file=1024)
+            with open(marker, "w") as f:
+                f.write(tag)
+    barrier()
+    return DATA_DIR
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch-size", type=int, default=212)  # ref deep_learning/2...py:342
+    ap.add_argument("--model", type=str, default="resnet18")
+    ap.add_argument("--rows", type=int, default=4096)
+    ap.add_argument("--workers-count", type=int, default=6)
+    ap.add_argument("--results-queue-size", type=int, default=8)
+    ap.add_argument("--bucket-cap-mb", type=int, default=32)
+    args = ap.parse_args()
+
+    use_cuda = torch.cuda.is_available()
+    image_hw = (224, 224)
+    batch = args.batch_size
+    if not use_cuda:  # CPU smoke config (no GPU in dev container)
+        image_hw, batch = (64, 64), 8
+        args.rows = min(args.rows, 512)
+
+    ctx = init_distributed()
+    n_gpus = ctx.world_size
+    device = ctx.device
+
+    data_dir = prepare_data(args.rows, image_hw, ctx.rank, n_gpus)
+
+    torch.manual_seed(1234)
+    model = ImageClassifier(args.model, num_classes=1000, lr=1e-5)
+    model.trainer = None
+    model.log = lambda *a, **k: None  # no metric plumbing in the bench loop
+    model.to(device)
+    if use_cuda:
+        model.to(memory_format=torch.channels_last)
+    runner = model
+    if n_gpus > 1:
+        kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,
+                      gradient_as_bucket_view=True)
+        if use_cuda:
+            kwargs["device_ids"] = [device.index]
+        runner = DDP(_TrainStepShim(model), **kwargs)
+    optimizer = model.configure_optimizers()
+
+    dm = ImageStreamDataModule(
+        data_dir, batch_size=batch,
+        workers_count=args.workers_count,
+        results_queue_size=args.results_queue_size,
+        cur_shard=ctx.rank if n_gpus > 1 else None,
+        shard_count=n_gpus if n_gpus > 1 else None,
+        image_hw=image_hw, device=device,
+        prefetch_depth=3,
+    )
+    loader = dm.train_dataloader()  # infinite reader
+    it = iter(loader)
+
+    amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                         enabled=use_cuda)
+
+    def one_step():
+        nonlocal it
+        b = next(it)
+        with amp:
+            if isinstance(runner, DDP):
+                loss = runner(b, 0)
+            else:
+                loss = model.training_step(b, 0)
+        optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    for _ in range(args.warmup):
+        one_step()
+
+    barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if ctx.backend == "nccl" else "cpu")
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    samples = n_gpus * batch * args.steps
+    value = samples / elapsed
+    if ctx.rank == 0:
+        print(json.dumps({
+            "metric": "samples/sec",
+            "value": value,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_cuda else "fp32-cpu-smoke",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * n_gpus,
+                "image": list(image_hw),
+                "parallelism": f"dp{n_gpus}",
+                "loader": {
+                    "workers_count": args.workers_count,
+                    "reader_pool_type": "thread",
+                    "results_queue_size": args.results_queue_size,
+                },
+            },
+        }))
+    dm.teardown()
+    destroy()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,176 @@
+"""DataLoader over a BatchReader, plus the device-side streaming stage.
+
+``DataLoader`` mirrors ``petastorm.pytorch.DataLoader(reader, batch_size)``
+(reference use: ``deep_learning/2.distributed-data-loading-petastorm.py:
+256-259``): slices variable-size row-group batches into fixed-size torch
+batches, carrying the remainder across row groups.
+
+``DeviceLoader`` is the MI355X replacement for Petastorm's "just call
+``.to(device)`` in the training loop" hidden cost: batches are staged into
+**pinned** host buffers (ring of ``depth``) and copied H2D with
+``hipMemcpyAsync`` on a dedicated **side stream**, so the copy of batch
+k+1 overlaps forward/backward of batch k. Consumer synchronizes via a
+recorded event, never a device-wide sync.
+"""
+from __future__ import annotations
+
+from typing import Dict, Iterator, Optional
+
+import numpy as np
+import torch
+
+from .reader import BatchReader
+
+_TORCH_DTYPE = {
+    np.dtype(np.float32): torch.float32,
+    np.dtype(np.float64): torch.float64,
+    np.dtype(np.float16): torch.float16,
+    np.dtype(np.int64): torch.int64,
+    np.dtype(np.int32): torch.int32,
+    np.dtype(np.int16): torch.int16,
+    np.dtype(np.uint8): torch.uint8,
+    np.dtype(np.int8): torch.int8,
+    np.dtype(np.bool_): torch.bool,
+}
+
+
+def _to_tensor(arr: np.ndarray) -> torch.Tensor:
+    if arr.dtype == np.dtype(np.uint16):
+        arr = arr.astype(np.int32)
+    return torch.from_numpy(np.ascontiguousarray(arr))
+
+
+class DataLoader:
+    """Fixed-batch-size iterator of ``dict[str, torch.Tensor]``."""
+
+    def __init__(self, reader: BatchReader, batch_size: int = 1,
+                 drop_last: bool = True):
+        self.reader = reader
+        self.batch_size = int(batch_size)
+        self.drop_last = drop_last
+        self._carry: Optional[Dict[str, np.ndarray]] = None
+
+    def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
+        carry: Optional[Dict[str, np.ndarray]] = None
+        for rg_batch in self.reader:
+            if carry is not None:
+                rg_batch = {
+                    k: np.concatenate([carry[k], rg_batch[k]], axis=0)
+                    for k in rg_batch
+                }
+                carry = None
+            n = len(next(iter(rg_batch.values())))
+            pos = 0
+            while n - pos >= self.batch_size:
+                yield {
+                    k: _to_tensor(v[pos:pos + self.batch_size])
+                    for k, v in rg_batch.items()
+                }
+                pos += self.batch_size
+            if pos < n:
+                carry = {k: v[pos:] for k, v in rg_batch.items()}
+        if carry is not None and not self.drop_last:
+            yield {k: _to_tensor(v) for k, v in carry.items()}
+
+    def close(self):
+        self.reader.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+        return False
+
+
+class DeviceLoader:
+    """Pinned-ring + side-stream H2D prefetcher wrapping a DataLoader.
+
+    The replacement for Petastorm's synchronous host→device copy: the
+    decode threads produce CPU batches; this stage owns ``depth`` pinned
+    slots and a non-default HIP stream. ``hipMemcpyAsync`` (issued by
+    ``Tensor.copy_(non_blocking=True)`` from pinned memory) for batch k+1
+    runs while the consumer computes on batch k; an event recorded on the
+    side stream is waited on by the compute stream — no global syncs.
+    """
+
+    def __init__(self, loader: DataLoader, device, depth: int = 2,
+                 dtype_map: Optional[Dict[str, torch.dtype]] = None):
+        self.loader = loader
+        self.device = torch.device(device)
+        self.depth = max(2, depth)
+        self.dtype_map = dtype_map or {}
+        self._use_cuda = self.device.type == "cuda"
+        if self._use_cuda:
+            self.copy_stream = torch.cuda.Stream(device=self.device)
+        self._pinned: Dict[int, Dict[str, torch.Tensor]] = {}
+        self._slot_events: Dict[int, torch.cuda.Event] = {}
+
+    def _pin_slot(self, slot: int, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        # A slot may be rewritten only after its previous H2D copy finished
+        # (the event is recorded on the copy stream right after the copy).
+        prev_ev = self._slot_events.get(slot)
+        if prev_ev is not None:
+            prev_ev.synchronize()
+        pinned = self._pinned.get(slot)
+        if pinned is None or any(
+            pinned[k].shape != v.shape or pinned[k].dtype != v.dtype
+            for k, v in batch.items()
+        ):
+            pinned = {
+                k: torch.empty_like(v, pin_memory=True) for k, v in batch.items()
+            }
+            self._pinned[slot] = pinned
+        for k, v in batch.items():
+            pinned[k].copy_(v)
+        return pinned
+
+    def __iter__(self):
+        if not self._use_cuda:
+            for batch in self.loader:
+                yield {
+                    k: v.to(self.dtype_map.get(k, v.dtype))
+                    for k, v in batch.items()
+                }
+            return
+
+        it = iter(self.loader)
+        inflight = []  # (event, device_batch)
+        slot = 0
+        exhausted = False
+        while True:
+            while not exhausted and len(inflight) < self.depth:
+                try:
+                    cpu_batch = next(it)
+                except StopIteration:
+                    exhausted = True
+                    break
+                pinned = self._pin_slot(slot, cpu_batch)
+                with torch.cuda.stream(self.copy_stream):
+                    dev = {}
+                    for k, v in pinned.items():
+                        d = v.to(self.device, non_blocking=True)
+                        want = self.dtype_map.get(k)
+                        if want is not None and d.dtype != want:
+                            d = d.to(want)
+                        dev[k] = d
+                    ev = torch.cuda.Event()
+                    ev.record(self.copy_stream)
+                self._slot_events[slot] = ev
+                slot = (slot + 1) % self.depth
+                inflight.append((ev, dev))
+            if not inflight:
+                return
+            ev, dev = inflight.pop(0)
+            torch.cuda.current_stream(self.device).wait_event(ev)
+            yield dev
+
+    def close(self):
+        self.loader.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+        return False

@@ -1,0 +1,30 @@
+"""HIP kernel bindings (gfx950).
+
+The compiled extension ``mi355x_scale.ops._C`` is built in-tree by
+``setup.py build_ext --inplace``. On a GPU box the HIP path is mandatory:
+ops raise if the extension is missing (so GPU tests can never silently
+fall back to eager PyTorch); on CPU the pure-torch reference
+implementations run (they are also the numerics references in tests).
+"""
+from __future__ import annotations
+
+import torch
+
+try:
+    from . import _C  # type: ignore
+    HAVE_EXT = True
+except ImportError:
+    _C = None
+    HAVE_EXT = False
+
+
+def require_ext() -> None:
+    if not HAVE_EXT:
+        raise RuntimeError(
+            "mi355x_scale.ops._C is not built — run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            "(the HIP path is mandatory on GPU; no silent eager fallback)"
+        )
+
+
+from .preprocess import normalize_images  # noqa: E402,F401

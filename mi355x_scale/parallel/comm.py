@@ -1,0 +1,123 @@
+"""Collective-communication bring-up: RCCL over xGMI.
+
+The reference delegates this entirely to Lightning DDP + NCCL and even
+*disables* P2P as a cloud workaround (``deep_learning/2.distributed-data-
+loading-petastorm.py:362-363`` sets ``NCCL_P2P_DISABLE=1``). On MI355X the
+xGMI mesh (7 point-to-point links per GPU, ≈153 GB/s each) is the whole
+point, so this module does the opposite: asserts peer access works on all
+pairs at startup, and never sets the disable knob.
+
+``init_distributed`` reads the torchrun env contract
+(RANK/LOCAL_RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT) and initializes the
+``nccl`` backend (which IS RCCL on ROCm) when a GPU is present, ``gloo``
+otherwise (CPU test path).
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+@dataclass
+class DistContext:
+    rank: int
+    world_size: int
+    local_rank: int
+    device: torch.device
+    backend: str
+
+    @property
+    def is_main(self) -> bool:
+        return self.rank == 0
+
+
+def env_rank() -> int:
+    return int(os.environ.get("RANK", "0"))
+
+
+def env_world_size() -> int:
+    return int(os.environ.get("WORLD_SIZE", "1"))
+
+
+def env_local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+
+
+def topology_probe(num_devices: Optional[int] = None) -> List[List[bool]]:
+    """All-pairs peer-access matrix (hipDeviceCanAccessPeer). On a healthy
+    MI355X node every off-diagonal entry is True (full xGMI mesh)."""
+    if not torch.cuda.is_available():
+        return []
+    n = num_devices or torch.cuda.device_count()
+    return [
+        [i != j and torch.cuda.can_device_access_peer(i, j) for j in range(n)]
+        for i in range(n)
+    ]
+
+
+def assert_xgmi_mesh() -> None:
+    """Fail loudly if any GPU pair lacks peer access — the inverse of the
+    reference's NCCL_P2P_DISABLE=1 workaround."""
+    if os.environ.get("NCCL_P2P_DISABLE") == "1":
+        raise RuntimeError(
+            "NCCL_P2P_DISABLE=1 is set: xGMI P2P is required on MI355X; "
+            "unset it (the reference's cloud workaround does not apply here)"
+        )
+    mat = topology_probe()
+    bad = [
+        (i, j)
+        for i, row in enumerate(mat)
+        for j, ok in enumerate(row)
+        if i != j and not ok
+    ]
+    if bad:
+        raise RuntimeError(f"peer access missing on GPU pairs: {bad}")
+
+
+def init_distributed(backend: Optional[str] = None,
+                     timeout_s: float = 600.0) -> DistContext:
+    rank, world, local = env_rank(), env_world_size(), env_local_rank()
+    use_cuda = torch.cuda.is_available()
+    if backend is None:
+        backend = "nccl" if use_cuda else "gloo"
+    if use_cuda:
+        torch.cuda.set_device(local % torch.cuda.device_count())
+        device = torch.device("cuda", local % torch.cuda.device_count())
+    else:
+        device = torch.device("cpu")
+    if world > 1 and not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group(
+            backend=backend, rank=rank, world_size=world,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+    if use_cuda and world > 1 and torch.cuda.device_count() > 1:
+        assert_xgmi_mesh()
+    return DistContext(rank=rank, world_size=world, local_rank=local,
+                       device=device, backend=backend)
+
+
+def barrier() -> None:
+    if dist.is_initialized():
+        if dist.get_backend() == "nccl":
+            dist.barrier(device_ids=[torch.cuda.current_device()])
+        else:
+            dist.barrier()
+
+
+def destroy() -> None:
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def all_reduce_mean(t: torch.Tensor) -> torch.Tensor:
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        t /= dist.get_world_size()
+    return t

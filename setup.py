@@ -1,0 +1,35 @@
+"""Build the in-tree HIP extensions for gfx950.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The resulting mi355x_scale/ops/_C*.so is git-ignored but travels with the
+repo snapshot to GPU boxes.
+"""
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+SRC = [
+    "mi355x_scale/ops/csrc/bindings.cpp",
+    "mi355x_scale/ops/csrc/preprocess.hip",
+]
+
+setup(
+    name="mi355x_scale",
+    version="0.1.0",
+    packages=["mi355x_scale"],
+    ext_modules=[
+        CUDAExtension(
+            name="mi355x_scale.ops._C",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)
