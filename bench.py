@@ -47,7 +47,7 @@ def prepare_data(rows: int, image_hw, rank: int, world: int) -> str:
             import shutil
             shutil.rmtree(DATA_DIR, ignore_errors=True)
             write_image_parquet(DATA_DIR, num_rows=rows, image_hw=image_hw,
-                                rows_per_group=64, rows_per_file=1024)
+                                rows_per_group=256, rows_per_file=1024)
             with open(marker, "w") as f:
                 f.write(tag)
     barrier()

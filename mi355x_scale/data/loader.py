@@ -37,7 +37,16 @@ _TORCH_DTYPE = {
 def _to_tensor(arr: np.ndarray) -> torch.Tensor:
     if arr.dtype == np.dtype(np.uint16):
         arr = arr.astype(np.int32)
-    return torch.from_numpy(np.ascontiguousarray(arr))
+    arr = np.ascontiguousarray(arr)
+    if not arr.flags.writeable:
+        # Zero-copy views of immutable arrow buffers: we only ever read
+        # from these tensors (they are staged into pinned buffers), so the
+        # "non-writable array" warning is noise.
+        import warnings
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            return torch.from_numpy(arr)
+    return torch.from_numpy(arr)
 
 
 class DataLoader:
@@ -133,37 +142,66 @@ class DeviceLoader:
                     for k, v in batch.items()
                 }
             return
+        yield from self._iter_cuda()
 
-        it = iter(self.loader)
-        inflight = []  # (event, device_batch)
-        slot = 0
-        exhausted = False
-        while True:
-            while not exhausted and len(inflight) < self.depth:
-                try:
-                    cpu_batch = next(it)
-                except StopIteration:
-                    exhausted = True
-                    break
-                pinned = self._pin_slot(slot, cpu_batch)
-                with torch.cuda.stream(self.copy_stream):
-                    dev = {}
-                    for k, v in pinned.items():
-                        d = v.to(self.device, non_blocking=True)
-                        want = self.dtype_map.get(k)
-                        if want is not None and d.dtype != want:
-                            d = d.to(want)
-                        dev[k] = d
-                    ev = torch.cuda.Event()
-                    ev.record(self.copy_stream)
-                self._slot_events[slot] = ev
-                slot = (slot + 1) % self.depth
-                inflight.append((ev, dev))
-            if not inflight:
-                return
-            ev, dev = inflight.pop(0)
-            torch.cuda.current_stream(self.device).wait_event(ev)
-            yield dev
+    def _iter_cuda(self):
+        """Batch assembly + pin-copy + async H2D all run on a background
+        thread (the HIP runtime is thread-safe); the consumer thread only
+        dequeues ready (event, device_batch) pairs and inserts a stream
+        wait — so host-side staging never blocks the training step."""
+        import queue as _q
+        import threading
+
+        out_q: "_q.Queue" = _q.Queue(maxsize=self.depth)
+        stop = threading.Event()
+        device, copy_stream = self.device, self.copy_stream
+
+        def _produce():
+            torch.cuda.set_device(device)
+            try:
+                slot = 0
+                for cpu_batch in self.loader:
+                    if stop.is_set():
+                        return
+                    pinned = self._pin_slot(slot, cpu_batch)
+                    with torch.cuda.stream(copy_stream):
+                        dev = {}
+                        for k, v in pinned.items():
+                            d = v.to(device, non_blocking=True)
+                            want = self.dtype_map.get(k)
+                            if want is not None and d.dtype != want:
+                                d = d.to(want)
+                            dev[k] = d
+                        ev = torch.cuda.Event()
+                        ev.record(copy_stream)
+                    self._slot_events[slot] = ev
+                    slot = (slot + 1) % self.depth
+                    while not stop.is_set():
+                        try:
+                            out_q.put((ev, dev), timeout=0.1)
+                            break
+                        except _q.Full:
+                            continue
+            finally:
+                while not stop.is_set():
+                    try:
+                        out_q.put(None, timeout=0.1)
+                        break
+                    except _q.Full:
+                        continue
+
+        t = threading.Thread(target=_produce, name="h2d-stager", daemon=True)
+        t.start()
+        try:
+            while True:
+                item = out_q.get()
+                if item is None:
+                    return
+                ev, dev = item
+                torch.cuda.current_stream(self.device).wait_event(ev)
+                yield dev
+        finally:
+            stop.set()
 
     def close(self):
         self.loader.close()

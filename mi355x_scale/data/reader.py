@@ -54,6 +54,7 @@ class BatchReader:
         num_epochs: Optional[int] = 1,
         shuffle_row_groups: bool = False,
         seed: Optional[int] = None,
+        arrow_transform=None,
     ):
         if reader_pool_type not in ("thread", "dummy"):
             raise ValueError(f"reader_pool_type must be 'thread' or 'dummy', got {reader_pool_type!r}")
@@ -62,6 +63,12 @@ class BatchReader:
         self.manifest = manifest
         self.schema_fields = schema_fields
         self.transform_spec = transform_spec
+        # arrow_transform: pyarrow.Table -> dict[str, np.ndarray]; the
+        # zero-copy fast path (skips pandas entirely). Mutually exclusive
+        # with transform_spec.
+        self.arrow_transform = arrow_transform
+        if arrow_transform is not None and transform_spec is not None:
+            raise ValueError("pass either transform_spec or arrow_transform")
         self.num_epochs = num_epochs
         self.workers_count = max(1, workers_count)
         self.reader_pool_type = reader_pool_type
@@ -121,6 +128,9 @@ class BatchReader:
     def _decode(self, ref: RowGroupRef) -> Dict[str, np.ndarray]:
         pf = self._parquet_file(ref.file_path)
         table = pf.read_row_group(ref.row_group, columns=self.schema_fields)
+        if self.arrow_transform is not None:
+            return {k: np.asarray(v)
+                    for k, v in self.arrow_transform(table).items()}
         pdf = table.to_pandas()
         if self.transform_spec is not None:
             pdf = self.transform_spec.apply(pdf)

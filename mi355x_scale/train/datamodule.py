@@ -33,6 +33,28 @@ def _bytes_to_nhwc(h: int, w: int):
     return _f
 
 
+def _arrow_images(h: int, w: int):
+    """Zero-copy arrow decode: the fixed_size_binary image column's data
+    buffer is viewed directly as uint8 [n,h,w,3] — no per-row python loop,
+    no np.stack memcpy. This is where the streaming path's host-side
+    headroom comes from (vs Petastorm's pandas round trip)."""
+    rb = h * w * 3
+
+    def _f(table):
+        col = table.column("image")
+        chunks = col.chunks if hasattr(col, "chunks") else [col]
+        views = []
+        for ch in chunks:
+            buf = ch.buffers()[1]
+            a = np.frombuffer(buf, dtype=np.uint8)
+            a = a[ch.offset * rb:(ch.offset + len(ch)) * rb]
+            views.append(a.reshape(len(ch), h, w, 3))
+        imgs = views[0] if len(views) == 1 else np.concatenate(views)
+        labels = table.column("label").to_numpy()
+        return {"image": imgs, "label": labels}
+    return _f
+
+
 class ImageStreamDataModule(DataModule):
     def __init__(
         self,
@@ -77,7 +99,7 @@ class ImageStreamDataModule(DataModule):
         h, w = self.image_hw
         reader = BatchReader(
             self._manifest,
-            transform_spec=TransformSpec(_bytes_to_nhwc(h, w)),
+            arrow_transform=_arrow_images(h, w),
             cur_shard=self.cur_shard,
             shard_count=self.shard_count,
             workers_count=self.workers_count,

@@ -99,6 +99,13 @@ class ImageClassifier(ScaleModule):
         return {"val_loss": loss, "val_accuracy": acc}
 
     def configure_optimizers(self):
+        if next(self.parameters()).is_cuda:
+            try:
+                # single fused kernel per step instead of a foreach chain
+                return torch.optim.Adam(self.parameters(), lr=self.lr,
+                                        fused=True)
+            except (RuntimeError, ValueError):
+                pass
         return torch.optim.Adam(self.parameters(), lr=self.lr)
 
 
