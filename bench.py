@@ -65,6 +65,12 @@ def main() -> None:
     ap.add_argument("--workers-count", type=int, default=6)
     ap.add_argument("--results-queue-size", type=int, default=8)
     ap.add_argument("--bucket-cap-mb", type=int, default=32)
+    ap.add_argument("--resident", action="store_true",
+                    help="skip the streaming loader; train on one resident "
+                         "device batch (isolates model step time)")
+    ap.add_argument("--timing", action="store_true",
+                    help="print per-step loader-wait vs compute breakdown "
+                         "to stderr")
     args = ap.parse_args()
 
     use_cuda = torch.cuda.is_available()
@@ -78,7 +84,9 @@ def main() -> None:
     n_gpus = ctx.world_size
     device = ctx.device
 
-    data_dir = prepare_data(args.rows, image_hw, ctx.rank, n_gpus)
+    data_dir = None
+    if not args.resident:
+        data_dir = prepare_data(args.rows, image_hw, ctx.rank, n_gpus)
 
     torch.manual_seed(1234)
     model = ImageClassifier(args.model, num_classes=1000, lr=1e-5)
@@ -96,24 +104,38 @@ def main() -> None:
         runner = DDP(_TrainStepShim(model), **kwargs)
     optimizer = model.configure_optimizers()
 
-    dm = ImageStreamDataModule(
-        data_dir, batch_size=batch,
-        workers_count=args.workers_count,
-        results_queue_size=args.results_queue_size,
-        cur_shard=ctx.rank if n_gpus > 1 else None,
-        shard_count=n_gpus if n_gpus > 1 else None,
-        image_hw=image_hw, device=device,
-        prefetch_depth=3,
-    )
-    loader = dm.train_dataloader()  # infinite reader
-    it = iter(loader)
+    if args.resident:
+        dm = None
+        g = torch.Generator().manual_seed(0)
+        fixed = {
+            "image": torch.randint(0, 256, (batch, *image_hw, 3),
+                                   dtype=torch.uint8, generator=g).to(device),
+            "label": torch.randint(0, 1000, (batch,), generator=g).to(device),
+        }
+        it = iter(lambda: fixed, None)  # infinite
+    else:
+        dm = ImageStreamDataModule(
+            data_dir, batch_size=batch,
+            workers_count=args.workers_count,
+            results_queue_size=args.results_queue_size,
+            cur_shard=ctx.rank if n_gpus > 1 else None,
+            shard_count=n_gpus if n_gpus > 1 else None,
+            image_hw=image_hw, device=device,
+            prefetch_depth=3,
+        )
+        loader = dm.train_dataloader()  # infinite reader
+        it = iter(loader)
 
     amp = torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                          enabled=use_cuda)
+    timing = {"wait": [], "step": []} if args.timing else None
 
     def one_step():
         nonlocal it
+        t0 = time.perf_counter() if timing is not None else 0.0
         b = next(it)
+        if timing is not None:
+            timing["wait"].append(time.perf_counter() - t0)
         with amp:
             if isinstance(runner, DDP):
                 loss = runner(b, 0)
@@ -122,6 +144,10 @@ def main() -> None:
         optimizer.zero_grad(set_to_none=True)
         loss.backward()
         optimizer.step()
+        if timing is not None:
+            if use_cuda:
+                torch.cuda.synchronize()
+            timing["step"].append(time.perf_counter() - t0)
         return loss
 
     for _ in range(args.warmup):
@@ -144,6 +170,16 @@ def main() -> None:
     if dist.is_initialized():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
+
+    if timing is not None:
+        import sys
+        import numpy as _np
+        w = _np.array(timing["wait"][args.warmup:]) * 1e3
+        s = _np.array(timing["step"][args.warmup:]) * 1e3
+        print(f"[timing rank {ctx.rank}] loader-wait ms p50={_np.median(w):.2f} "
+              f"p95={_np.percentile(w,95):.2f} max={w.max():.2f} | "
+              f"full-step ms p50={_np.median(s):.2f} p95={_np.percentile(s,95):.2f} "
+              f"max={s.max():.2f}", file=sys.stderr)
 
     samples = n_gpus * batch * args.steps
     value = samples / elapsed
@@ -173,7 +209,8 @@ def main() -> None:
                 },
             },
         }))
-    dm.teardown()
+    if dm is not None:
+        dm.teardown()
     destroy()
 
 

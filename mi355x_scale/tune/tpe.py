@@ -1,0 +1,148 @@
+"""Tree-structured Parzen Estimator (TPE) suggestion algorithm.
+
+Native re-implementation of the hyperopt ``tpe.suggest`` role the
+reference uses everywhere (``hyperopt/1. hyperopt.py:95``,
+``group_apply/02_Fine_Grained_Demand_Forecasting.py:311``): factorized
+1-D adaptive Parzen estimators per dimension — observations are split
+into a "good" fraction (lowest-loss γ quantile) and the rest; candidates
+are drawn from the good-density l(x) and ranked by l(x)/g(x)
+(equivalently EI). Bandwidths follow the adjacent-point heuristic.
+
+This is a faithful algorithmic re-implementation, not a bit-for-bit port:
+seeded runs are deterministic here, but do not reproduce hyperopt's exact
+draws.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Sequence, Tuple
+
+import numpy as np
+
+from .space import Expr, flatten_space
+
+N_STARTUP_TRIALS = 10     # random exploration before TPE kicks in
+N_EI_CANDIDATES = 24      # candidates drawn from l(x) per dimension
+GAMMA = 0.25              # good/bad split quantile
+PRIOR_WEIGHT = 1.0
+
+
+def _split(losses: np.ndarray) -> int:
+    """Number of observations in the 'good' set (hyperopt-style)."""
+    n = len(losses)
+    return max(1, min(int(math.ceil(GAMMA * math.sqrt(n) * 4)), n - 1, 25))
+
+
+def _adaptive_parzen(mus: np.ndarray, prior_mu: float, prior_sigma: float
+                     ) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Weights, means, sigmas of the 1-D Parzen mixture over ``mus`` plus
+    the prior component (hyperopt's adaptive_parzen_normal heuristic)."""
+    order = np.argsort(mus)
+    smus = mus[order]
+    m = len(smus)
+    # insert prior as an extra component
+    means = np.concatenate([smus, [prior_mu]])
+    means_sorted = np.sort(means)
+    sigmas = np.empty(m + 1)
+    for i, mu in enumerate(means):
+        pos = np.searchsorted(means_sorted, mu)
+        left = means_sorted[pos - 1] if pos > 0 else mu - prior_sigma
+        right = means_sorted[pos + 1] if pos + 1 < len(means_sorted) else mu + prior_sigma
+        sigmas[i] = max(abs(mu - left), abs(right - mu))
+    # clamp
+    minsigma = prior_sigma / max(100.0, (1.0 + m))
+    sigmas = np.clip(sigmas, minsigma, prior_sigma)
+    sigmas[m] = prior_sigma  # prior keeps its width
+    weights = np.ones(m + 1)
+    weights[m] = PRIOR_WEIGHT
+    if m > 20:  # downweight old points linearly (hyperopt ramp)
+        ramp = np.linspace(1.0 / m, 1.0, m)
+        weights[:m] = ramp
+    weights /= weights.sum()
+    return weights, means, sigmas
+
+
+def _gmm_logpdf(x: np.ndarray, w: np.ndarray, mu: np.ndarray,
+                sigma: np.ndarray) -> np.ndarray:
+    x = x[:, None]
+    z = (x - mu[None, :]) / sigma[None, :]
+    comp = -0.5 * z * z - np.log(sigma[None, :] * math.sqrt(2 * math.pi))
+    comp += np.log(w[None, :])
+    mx = comp.max(axis=1, keepdims=True)
+    return (mx + np.log(np.exp(comp - mx).sum(axis=1, keepdims=True))).ravel()
+
+
+def _gmm_sample(rng: np.random.Generator, n: int, w: np.ndarray,
+                mu: np.ndarray, sigma: np.ndarray) -> np.ndarray:
+    idx = rng.choice(len(w), size=n, p=w)
+    return rng.normal(mu[idx], sigma[idx])
+
+
+def _prior_mu_sigma(node: Expr) -> Tuple[float, float]:
+    from .space import (Choice, IntCast, LogNormal, LogUniform, Normal,
+                        QLogUniform, QUniform, RandInt, Uniform)
+    if isinstance(node, IntCast):
+        return _prior_mu_sigma(node.inner)
+    if isinstance(node, (Uniform, QUniform)):
+        return (node.low + node.high) / 2.0, (node.high - node.low)
+    if isinstance(node, (LogUniform, QLogUniform)):
+        return (node.low + node.high) / 2.0, (node.high - node.low)
+    if isinstance(node, (Normal, LogNormal)):
+        return node.mu, node.sigma
+    if isinstance(node, RandInt):
+        return (node.upper - 1) / 2.0, max(node.upper / 2.0, 1.0)
+    if isinstance(node, Choice):
+        k = len(node.options)
+        return (k - 1) / 2.0, max(k / 2.0, 1.0)
+    raise TypeError(f"unsupported node {node!r}")
+
+
+class TPE:
+    def __init__(self, n_startup_trials: int = N_STARTUP_TRIALS,
+                 n_ei_candidates: int = N_EI_CANDIDATES):
+        self.n_startup = n_startup_trials
+        self.n_ei = n_ei_candidates
+
+    def propose(self, space, history: Sequence[Tuple[Dict, float]],
+                rng: np.random.Generator) -> Dict:
+        """history: [(params_dict, loss)] of COMPLETED ok trials."""
+        nodes = flatten_space(space)
+        if len(history) < self.n_startup:
+            return {lbl: node.sample(rng) for lbl, node in nodes.items()}
+        losses = np.array([l for _, l in history], dtype=np.float64)
+        n_good = _split(losses)
+        good_idx = np.argsort(losses)[:n_good]
+        good_set = set(good_idx.tolist())
+        out = {}
+        for lbl, node in nodes.items():
+            obs = np.array([node.to_internal(p[lbl]) for p, _ in history])
+            prior_mu, prior_sigma = _prior_mu_sigma(node)
+            gw, gm, gs = _adaptive_parzen(
+                np.array([obs[i] for i in range(len(obs)) if i in good_set]),
+                prior_mu, prior_sigma)
+            bw, bm, bs = _adaptive_parzen(
+                np.array([obs[i] for i in range(len(obs)) if i not in good_set]),
+                prior_mu, prior_sigma)
+            cand = _gmm_sample(rng, self.n_ei, gw, gm, gs)
+            cand = np.array([node.clip_internal(c) for c in cand])
+            score = _gmm_logpdf(cand, gw, gm, gs) - _gmm_logpdf(cand, bw, bm, bs)
+            best = cand[int(np.argmax(score))]
+            out[lbl] = node.from_internal(best)
+        return out
+
+
+class _RandomAlgo:
+    def propose(self, space, history, rng):
+        return {lbl: node.sample(rng)
+                for lbl, node in flatten_space(space).items()}
+
+
+class _AlgoNamespace:
+    """hyperopt-compat markers: pass ``tpe.suggest`` / ``rand.suggest``
+    as the ``algo=`` argument of fmin."""
+    def __init__(self, factory):
+        self.suggest = factory
+
+
+tpe = _AlgoNamespace(TPE)
+rand = _AlgoNamespace(_RandomAlgo)
