@@ -156,14 +156,27 @@ class DeviceLoader:
         stop = threading.Event()
         device, copy_stream = self.device, self.copy_stream
 
+        import os
+        import time as _time
+        debug = os.environ.get("MI355X_LOADER_DEBUG") == "1"
+        stats = {"dequeue": [], "pin": [], "h2d_issue": []} if debug else None
+
         def _produce():
             torch.cuda.set_device(device)
             try:
                 slot = 0
-                for cpu_batch in self.loader:
+                it = iter(self.loader)
+                while True:
+                    t0 = _time.perf_counter()
+                    try:
+                        cpu_batch = next(it)
+                    except StopIteration:
+                        return
                     if stop.is_set():
                         return
+                    t1 = _time.perf_counter()
                     pinned = self._pin_slot(slot, cpu_batch)
+                    t2 = _time.perf_counter()
                     with torch.cuda.stream(copy_stream):
                         dev = {}
                         for k, v in pinned.items():
@@ -174,6 +187,11 @@ class DeviceLoader:
                             dev[k] = d
                         ev = torch.cuda.Event()
                         ev.record(copy_stream)
+                    if debug:
+                        t3 = _time.perf_counter()
+                        stats["dequeue"].append(t1 - t0)
+                        stats["pin"].append(t2 - t1)
+                        stats["h2d_issue"].append(t3 - t2)
                     self._slot_events[slot] = ev
                     slot = (slot + 1) % self.depth
                     while not stop.is_set():
@@ -183,6 +201,15 @@ class DeviceLoader:
                         except _q.Full:
                             continue
             finally:
+                if debug and stats["dequeue"]:
+                    import numpy as _np
+                    import sys
+                    for k, v in stats.items():
+                        a = _np.array(v) * 1e3
+                        print(f"[loader-debug] {k}: p50={_np.median(a):.2f} "
+                              f"p95={_np.percentile(a, 95):.2f} "
+                              f"max={a.max():.2f} n={len(a)}",
+                              file=sys.stderr)
                 while not stop.is_set():
                     try:
                         out_q.put(None, timeout=0.1)

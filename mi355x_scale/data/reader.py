@@ -91,6 +91,7 @@ class BatchReader:
         self._done_workers = 0
         self._done_lock = threading.Lock()
         self._pf_cache: Dict[str, pq.ParquetFile] = {}
+        self._decode_times: List[float] = []
         self._pf_lock = threading.Lock()
         if reader_pool_type == "thread":
             for i in range(self.workers_count):
@@ -150,12 +151,18 @@ class BatchReader:
 
     # -- worker loop -------------------------------------------------------
     def _worker(self) -> None:
+        import os
+        import time as _time
+        debug = os.environ.get("MI355X_LOADER_DEBUG") == "1"
         try:
             while not self._stop.is_set():
                 ref = self._next_ref()
                 if ref is None:
                     break
+                t0 = _time.perf_counter() if debug else 0.0
                 batch = self._decode(ref)
+                if debug:
+                    self._decode_times.append(_time.perf_counter() - t0)
                 while not self._stop.is_set():
                     try:
                         self._results.put(batch, timeout=0.1)
@@ -208,6 +215,14 @@ class BatchReader:
     def close(self) -> None:
         self.stop()
         self.join()
+        if self._decode_times:
+            import sys
+            import numpy as _np
+            a = _np.array(self._decode_times) * 1e3
+            print(f"[reader-debug] decode ms: p50={_np.median(a):.2f} "
+                  f"p95={_np.percentile(a, 95):.2f} max={a.max():.2f} "
+                  f"n={len(a)}", file=sys.stderr)
+            self._decode_times = []
 
     def __enter__(self):
         return self
