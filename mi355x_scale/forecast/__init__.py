@@ -1,1 +1,8 @@
-"""(placeholder — populated in later milestones)"""
+"""W1: demand forecasting — SARIMAX-lite (CPU reference), per-SKU tuning
+pipeline, and the batched CDNA4 GPU fit."""
+
+from .sarimax import SARIMAX, SARIMAXResults  # noqa: F401
+from .pipeline import (add_exo_variables, split_train_score_data,  # noqa: F401
+                       evaluate_model, build_tune_and_score_model,
+                       run_fine_grained_forecast, TUNING_SCHEMA,
+                       FORECAST_HORIZON, SEARCH_SPACE)

@@ -1,0 +1,130 @@
+"""W1 per-SKU demand-forecast pipeline.
+
+Native equivalent of the reference's notebook functions
+(``group_apply/02_Fine_Grained_Demand_Forecasting.py``):
+
+  * ``add_exo_variables``      (ref :343-358) — covid/christmas/new_year
+    exogenous indicator columns from the Date column.
+  * ``split_train_score_data`` (ref :372-380) — last ``forecast_horizon``
+    weeks held out for scoring.
+  * ``evaluate_model``         (ref :264-282) — fit candidate (p,d,q),
+    MSE on the held-out window.
+  * ``build_tune_and_score_model`` (ref :417-494) — per-group: sort by
+    date, split, nested sequential TPE over (p,d,q) ∈ [0,4]×[0,2]×[0,4]
+    with ``max_evals=10`` (ref :461-469), refit best on train+score,
+    one-step-ahead fit over the full range, return the forecast frame.
+
+Used two ways: through the ``groupby`` engine (pandas path, config 1) and
+as the semantics reference for the batched GPU fit (``batched.py``).
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import numpy as np
+import pandas as pd
+
+from ..tune import Trials, fmin, hp, scope, tpe
+from .sarimax import SARIMAX
+
+FORECAST_HORIZON = 40  # weeks, ref :341
+EXO_COLS = ["covid", "christmas", "new_year"]
+
+TUNING_SCHEMA = ("Product string, SKU string, Date date, Demand float, "
+                 "Demand_Fitted float")
+
+SEARCH_SPACE = {
+    "p": scope.int(hp.quniform("p", 0, 4, 1)),
+    "d": scope.int(hp.quniform("d", 0, 2, 1)),
+    "q": scope.int(hp.quniform("q", 0, 4, 1)),
+}
+
+
+def add_exo_variables(pdf: pd.DataFrame) -> pd.DataFrame:
+    """Compute the three exogenous indicators from Date (ref :343-358)."""
+    out = pdf.copy()
+    date = pd.to_datetime(out["Date"])
+    covid_start = pd.Timestamp("2020-03-01")
+    weeks_since = ((date - covid_start).dt.days // 7)
+    in_covid = (weeks_since >= 0) & (weeks_since < 16)
+    out["covid"] = np.where(in_covid,
+                            np.exp(-0.25 * weeks_since.clip(lower=0)), 0.0)
+    out["christmas"] = ((date.dt.month == 12) & (date.dt.day >= 18)).astype(float)
+    out["new_year"] = ((date.dt.month == 1) & (date.dt.day <= 7)).astype(float)
+    return out
+
+
+def split_train_score_data(pdf: pd.DataFrame,
+                           horizon: int = FORECAST_HORIZON
+                           ) -> Tuple[pd.DataFrame, pd.DataFrame]:
+    """Last ``horizon`` rows are the score window (ref :372-380)."""
+    return pdf.iloc[:-horizon], pdf.iloc[-horizon:]
+
+
+def evaluate_model(params: Dict, train: pd.DataFrame, score: pd.DataFrame,
+                   exo_cols=EXO_COLS) -> Dict:
+    """Fit (p,d,q) on train, forecast the score window, return MSE loss
+    (ref evaluate_model :435-459)."""
+    from ..tune import STATUS_FAIL, STATUS_OK
+    p, d, q = int(params["p"]), int(params["d"]), int(params["q"])
+    try:
+        res = SARIMAX(train["Demand"].to_numpy(),
+                      exog=train[exo_cols].to_numpy(),
+                      order=(p, d, q)).fit(disp=False)
+        fc = res.forecast(len(score), exog=score[exo_cols].to_numpy())
+        mse = float(np.mean((score["Demand"].to_numpy() - fc) ** 2))
+        if not np.isfinite(mse):
+            return {"loss": None, "status": STATUS_FAIL}
+        return {"loss": mse, "status": STATUS_OK}
+    except Exception:
+        return {"loss": None, "status": STATUS_FAIL}
+
+
+def build_tune_and_score_model(sku_pdf: pd.DataFrame,
+                               horizon: int = FORECAST_HORIZON,
+                               max_evals: int = 10,
+                               seed: int = 123) -> pd.DataFrame:
+    """The applyInPandas group function (ref :417-494): returns a frame
+    matching TUNING_SCHEMA with one-step-ahead fitted demand."""
+    pdf = sku_pdf.sort_values("Date").reset_index(drop=True)
+    train, score = split_train_score_data(pdf, horizon)
+
+    trials = Trials()
+    fmin(lambda prm: evaluate_model(prm, train, score),
+         SEARCH_SPACE, algo=tpe.suggest, max_evals=max_evals,
+         trials=trials, rstate=np.random.default_rng(seed))
+    bt = trials.best_trial["misc"]["vals"]
+    p, d, q = int(bt["p"][0]), int(bt["d"][0]), int(bt["q"][0])
+
+    # final refit on the whole series (ref :472-481)
+    res = SARIMAX(pdf["Demand"].to_numpy(),
+                  exog=pdf[EXO_COLS].to_numpy(),
+                  order=(p, d, q)).fit(disp=False)
+    fitted = res.predict(0, len(pdf) - 1)
+    return pd.DataFrame({
+        "Product": pdf["Product"],
+        "SKU": pdf["SKU"],
+        "Date": pd.to_datetime(pdf["Date"]),
+        "Demand": pdf["Demand"].astype(float),
+        "Demand_Fitted": fitted,
+    })
+
+
+def run_fine_grained_forecast(demand_df: pd.DataFrame,
+                              num_workers: Optional[int] = None,
+                              max_evals: int = 10,
+                              horizon: int = FORECAST_HORIZON
+                              ) -> pd.DataFrame:
+    """The whole W1 job: enrich → group → tune+score per SKU (the
+    reference's :520-528 chain) on the local process pool."""
+    from ..groupby import LocalFrame
+    if not set(EXO_COLS) <= set(demand_df.columns):
+        demand_df = add_exo_variables(demand_df)
+    n_groups = demand_df.groupby(["Product", "SKU"]).ngroups
+    frame = LocalFrame(demand_df, num_workers=num_workers)
+    return (frame.repartition(n_groups, "Product", "SKU")
+                 .groupBy("Product", "SKU")
+                 .applyInPandas(
+                     lambda g: build_tune_and_score_model(
+                         g, horizon=horizon, max_evals=max_evals),
+                     TUNING_SCHEMA))

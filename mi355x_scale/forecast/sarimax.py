@@ -245,6 +245,17 @@ class SARIMAX:
                 X2, y2 = _build_lag_matrix(u, eps, p, q, m)
                 c2 = _ols(X2, y2)
                 phi, theta = c2[:p], c2[p:]
+                # Shrink toward stationarity/invertibility: the ê and
+                # forecast recursions diverge when Σ|φ| or Σ|θ| > 1
+                # (statsmodels hides this inside enforce_stationarity;
+                # here the fixed-schedule estimator clips instead — same
+                # rule as the GPU kernel).
+                sp = np.sum(np.abs(phi))
+                if sp > 0.98:
+                    phi *= 0.98 / sp
+                st = np.sum(np.abs(theta))
+                if st > 0.98:
+                    theta *= 0.98 / st
                 # recompute innovations under (phi, theta)
                 new_eps = np.zeros(T)
                 for t in range(T):

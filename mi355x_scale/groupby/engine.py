@@ -83,6 +83,14 @@ def _run_chunk(fn: Callable, groups: List[Tuple[tuple, pd.DataFrame]],
     return outs, fails
 
 
+def _run_chunk_pickled(payload: bytes):
+    # fn is shipped with cloudpickle so lambdas/closures work (the same
+    # serialization contract Spark gives applyInPandas UDFs).
+    import cloudpickle
+    fn, groups, schema = cloudpickle.loads(payload)
+    return _run_chunk(fn, groups, schema)
+
+
 class GroupedFrame:
     def __init__(self, df: pd.DataFrame, keys: Sequence[str],
                  num_workers: Optional[int] = None,
@@ -108,10 +116,12 @@ class GroupedFrame:
                       for i in range(0, len(groups), chunk)]
             outs, self.failures = [], []
             import multiprocessing as mp
+            import cloudpickle
             with ProcessPoolExecutor(
                     max_workers=n_workers,
                     mp_context=mp.get_context("spawn")) as pool:
-                futs = [pool.submit(_run_chunk, fn, c, schema)
+                futs = [pool.submit(_run_chunk_pickled,
+                                    cloudpickle.dumps((fn, c, schema)))
                         for c in chunks]
                 for f in futs:
                     o, e = f.result()

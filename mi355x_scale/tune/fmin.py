@@ -110,6 +110,15 @@ def _worker_init(device_queue):
         os.environ["CUDA_VISIBLE_DEVICES"] = str(dev)
 
 
+def _eval_trial_pickled(fn_blob: bytes, args_blob: bytes) -> Dict:
+    """Pool entry: fn ships via cloudpickle so closures/lambdas work (the
+    serialization contract SparkTrials gives objectives)."""
+    import cloudpickle
+    fn = cloudpickle.loads(fn_blob)
+    args = cloudpickle.loads(args_blob)
+    return _eval_trial(fn, args)
+
+
 def _eval_trial(fn: Callable, args: Any) -> Dict:
     try:
         out = fn(args)
@@ -163,7 +172,9 @@ def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
             result = _eval_trial(fn, bind_params(space, params))
             _record(tid, params, result)
     else:
+        import cloudpickle
         import multiprocessing as mp
+        fn_blob = cloudpickle.dumps(fn)
         ctx = mp.get_context("spawn")
         dq = ctx.Queue()
         n_dev = par
@@ -181,8 +192,9 @@ def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
             while done_n < max_evals:
                 while len(pending) < par and issued < max_evals:
                     params = optimizer.propose(space, history, rng)
-                    fut = pool.submit(_eval_trial, fn,
-                                      bind_params(space, params))
+                    fut = pool.submit(
+                        _eval_trial_pickled, fn_blob,
+                        cloudpickle.dumps(bind_params(space, params)))
                     pending[fut] = (issued, params)
                     issued += 1
                 done, _ = wait(list(pending), return_when=FIRST_COMPLETED)
