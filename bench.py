@@ -65,6 +65,8 @@ def main() -> None:
     ap.add_argument("--workers-count", type=int, default=6)
     ap.add_argument("--results-queue-size", type=int, default=8)
     ap.add_argument("--bucket-cap-mb", type=int, default=32)
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph step capture (eager DDP path)")
     ap.add_argument("--resident", action="store_true",
                     help="skip the streaming loader; train on one resident "
                          "device batch (isolates model step time)")
@@ -95,14 +97,19 @@ def main() -> None:
     model.to(device)
     if use_cuda:
         model.to(memory_format=torch.channels_last)
+    use_graph = use_cuda and not args.no_graph
     runner = model
-    if n_gpus > 1:
-        kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,
-                      gradient_as_bucket_view=True)
-        if use_cuda:
-            kwargs["device_ids"] = [device.index]
-        runner = DDP(_TrainStepShim(model), **kwargs)
-    optimizer = model.configure_optimizers()
+    if use_graph:
+        optimizer = torch.optim.Adam(model.parameters(), lr=1e-5,
+                                     fused=True, capturable=True)
+    else:
+        if n_gpus > 1:
+            kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,
+                          gradient_as_bucket_view=True)
+            if use_cuda:
+                kwargs["device_ids"] = [device.index]
+            runner = DDP(_TrainStepShim(model), **kwargs)
+        optimizer = model.configure_optimizers()
 
     if args.resident:
         dm = None
@@ -130,20 +137,30 @@ def main() -> None:
                          enabled=use_cuda)
     timing = {"wait": [], "step": []} if args.timing else None
 
+    graphed = None
+    if use_graph:
+        from mi355x_scale.train.graphstep import GraphedTrainStep
+        example = next(it)
+        graphed = GraphedTrainStep(model, optimizer, example,
+                                   world_size=n_gpus, warmup=3)
+
     def one_step():
         nonlocal it
         t0 = time.perf_counter() if timing is not None else 0.0
         b = next(it)
         if timing is not None:
             timing["wait"].append(time.perf_counter() - t0)
-        with amp:
-            if isinstance(runner, DDP):
-                loss = runner(b, 0)
-            else:
-                loss = model.training_step(b, 0)
-        optimizer.zero_grad(set_to_none=True)
-        loss.backward()
-        optimizer.step()
+        if graphed is not None:
+            loss = graphed.step(b)
+        else:
+            with amp:
+                if isinstance(runner, DDP):
+                    loss = runner(b, 0)
+                else:
+                    loss = model.training_step(b, 0)
+            optimizer.zero_grad(set_to_none=True)
+            loss.backward()
+            optimizer.step()
         if timing is not None:
             if use_cuda:
                 torch.cuda.synchronize()
@@ -201,7 +218,7 @@ def main() -> None:
                 "model": args.model,
                 "global_batch": batch * n_gpus,
                 "image": list(image_hw),
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": f"dp{n_gpus}" + ("+hipgraph" if use_graph else ""),
                 "loader": {
                     "workers_count": args.workers_count,
                     "reader_pool_type": "thread",

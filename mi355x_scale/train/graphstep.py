@@ -1,0 +1,103 @@
+"""hipGraph-captured training step.
+
+The profiled streaming bench showed the step is launch-bound: ~400 kernel
+launches per step from a GIL-contended Python thread leave the GPU ~45%
+idle (p50 15.5 ms, p95 80+ ms). The MI355X-idiomatic fix (north star:
+"HIP streams and graphs instead of a tracing compiler") is to capture the
+whole step — grad-zero, forward, loss, backward, (optimizer) — into one
+hipGraph and replay it with a single launch per step.
+
+Data parallism is done manually around the graph (DDP's bucketed hooks
+can't fire inside a replay): gradients live in ONE flat buffer
+(``p.grad`` are views), the all-reduce is a single RCCL call on that
+buffer between the fwd/bwd graph and the optimizer graph. For ResNet-18
+(~45 MB fp32 grads) one flat all-reduce over xGMI beats DDP's bucket
+pipeline — there is almost no backward left to overlap with after graph
+capture.
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class GraphedTrainStep:
+    def __init__(self, model, optimizer, example_batch: Dict[str, torch.Tensor],
+                 autocast_dtype: Optional[torch.dtype] = torch.bfloat16,
+                 world_size: int = 1, warmup: int = 3):
+        self.model = model
+        self.optimizer = optimizer
+        self.world_size = world_size
+        self.device = next(model.parameters()).device
+        assert self.device.type == "cuda", "graph capture needs a GPU"
+        self.autocast_dtype = autocast_dtype
+
+        # Static input buffers (graph replays read these addresses).
+        self.static_batch = {
+            k: v.to(self.device).clone() for k, v in example_batch.items()
+        }
+
+        # Flat gradient buffer; every p.grad is a view into it so the
+        # cross-rank all-reduce is ONE call.
+        params = [p for p in model.parameters() if p.requires_grad]
+        total = sum(p.numel() for p in params)
+        self.flat_grads = torch.zeros(total, device=self.device,
+                                      dtype=torch.float32)
+        off = 0
+        for p in params:
+            p.grad = self.flat_grads[off:off + p.numel()].view_as(p)
+            off += p.numel()
+
+        def _fwd_bwd():
+            self.flat_grads.zero_()
+            with torch.autocast(device_type="cuda", dtype=autocast_dtype,
+                                enabled=autocast_dtype is not None):
+                loss = model.training_step(self.static_batch, 0)
+            loss.backward()
+            return loss
+
+        # Warmup on a side stream (MIOpen find, allocator steady state)
+        s = torch.cuda.Stream(device=self.device)
+        s.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(s):
+            for _ in range(max(1, warmup)):
+                _fwd_bwd()
+                if world_size > 1:
+                    self.flat_grads.div_(world_size)
+                optimizer.step()
+        torch.cuda.current_stream(self.device).wait_stream(s)
+        torch.cuda.synchronize(self.device)
+
+        # Capture forward+backward (one graph)...
+        self.g_fwd_bwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_fwd_bwd):
+            self.static_loss = _fwd_bwd()
+
+        # ...and the optimizer step (second graph; the eager RCCL
+        # all-reduce slots between the two replays when world_size > 1).
+        self.g_opt = None
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                if world_size > 1:
+                    self.flat_grads.div_(world_size)
+                optimizer.step()
+            self.g_opt = g
+        except RuntimeError:
+            self.g_opt = None  # non-capturable optimizer: step eagerly
+
+    def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        for k, v in batch.items():
+            self.static_batch[k].copy_(v, non_blocking=True)
+        self.g_fwd_bwd.replay()
+        if self.world_size > 1:
+            dist.all_reduce(self.flat_grads)
+        if self.g_opt is not None:
+            self.g_opt.replay()
+        else:
+            if self.world_size > 1:
+                self.flat_grads.div_(self.world_size)
+            self.optimizer.step()
+        return self.static_loss
