@@ -101,7 +101,7 @@ def main() -> None:
     runner = model
     if use_graph:
         optimizer = torch.optim.Adam(model.parameters(), lr=1e-5,
-                                     fused=True, capturable=True)
+                                     foreach=True, capturable=True)
     else:
         if n_gpus > 1:
             kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,

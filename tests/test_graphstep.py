@@ -37,8 +37,7 @@ def test_graphed_step_matches_eager():
     # graphed (warmup batches are extra optimizer steps, so warm up with
     # a throwaway copy of the state then restore)
     state0 = copy.deepcopy(model_g.state_dict())
-    opt_g = torch.optim.Adam(model_g.parameters(), lr=1e-3, fused=True,
-                             capturable=True)
+    opt_g = torch.optim.Adam(model_g.parameters(), lr=1e-3, foreach=True, capturable=True)
     gs = GraphedTrainStep(model_g, opt_g, batches[0], world_size=1, warmup=2)
     model_g.load_state_dict(state0)
     # reset optimizer state mutated by warmup
@@ -65,8 +64,7 @@ def test_graphed_step_loss_finite_and_changing():
     torch.manual_seed(0)
     dev = torch.device("cuda:0")
     model = ImageClassifier("resnet18", num_classes=10, lr=1e-2).to(dev)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-2, fused=True,
-                           capturable=True)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-2, foreach=True, capturable=True)
     b = {"image": torch.randint(0, 256, (8, 64, 64, 3), dtype=torch.uint8,
                                 device=dev),
          "label": torch.randint(0, 10, (8,), device=dev)}
