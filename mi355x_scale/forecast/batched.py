@@ -78,10 +78,9 @@ def _levinson(r: np.ndarray, M: int) -> np.ndarray:
     for k in range(1, M + 1):
         acc = r[k] - np.dot(a[:k - 1], r[k - 1:0:-1])
         lam = acc / e
-        a_new = a.copy()
-        a_new[k - 1] = lam
-        a_new[:k - 1] = a[:k - 1] - lam * a[k - 2::-1]
-        a = a_new
+        prev = a[:k - 1].copy()
+        a[k - 1] = lam
+        a[:k - 1] = prev - lam * prev[::-1]
         e *= (1.0 - lam * lam)
         if e <= 0:
             e = 1e-12
@@ -209,6 +208,70 @@ def batched_fit_reference(
             fc = _forecast_y(yg, S, H, d, wm, reg_future, phi, theta, u, eps)
             mse[g, ci] = np.mean((yg[S:] - fc) ** 2)
     return mse
+
+
+def _designs_to_gpu(designs: List[ExogDesign], device):
+    import torch
+    xs, ps = [], []
+    for dz in sorted(designs, key=lambda z: z.d):
+        xs.append(torch.tensor(dz.Xc_full, dtype=torch.float32,
+                               device=device).contiguous())
+        ps.append(torch.tensor(dz.P, dtype=torch.float32,
+                               device=device).contiguous())
+    return xs, ps
+
+
+def batched_eval_gpu(y, exog, orders, train_len: int, device="cuda"):
+    """GPU candidate evaluation: returns (mse [G,C] torch.f32, status
+    [G,C] torch.u8). ``y`` is [G,T] (numpy or torch)."""
+    import torch
+    from ..ops import require_ext, _C
+    require_ext()
+    y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32)
+    G, T = y_t.shape
+    yT = y_t.t().contiguous().to(device)
+    designs = make_exog_designs(exog, train_len)
+    xs, ps = _designs_to_gpu(designs, device)
+    orders_t = torch.tensor(list(orders), dtype=torch.int32,
+                            device=device).reshape(-1, 3).contiguous()
+    C = orders_t.shape[0]
+    mse = torch.empty((C, G), dtype=torch.float32, device=device)
+    status = torch.empty((C, G), dtype=torch.uint8, device=device)
+    _C.groupfit_eval(yT, xs[0], xs[1], xs[2], ps[0], ps[1], ps[2],
+                     orders_t, mse, status, train_len)
+    return mse.t().contiguous(), status.t().contiguous()
+
+
+def batched_fit_gpu(y, exog, orders, train_len: int, device="cuda"):
+    """Full W1 GPU pipeline: evaluate candidates, pick the best per group,
+    final-fit on the whole series. Returns a dict with ``best_order``
+    [G,3], ``mse`` [G,C], ``fitted`` [G,T], ``params`` [G,1+KX+8],
+    ``status`` [G] (all torch tensors on ``device``)."""
+    import torch
+    from ..ops import require_ext, _C
+    require_ext()
+    y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32)
+    G, T = y_t.shape
+    yT = y_t.t().contiguous().to(device)
+    mse, status = batched_eval_gpu(y, exog, orders, train_len, device)
+    orders_t = torch.tensor(list(orders), dtype=torch.int32, device=device
+                            ).reshape(-1, 3)
+    best_ci = mse.argmin(dim=1)                        # [G]
+    best_order = orders_t[best_ci].contiguous()        # [G,3]
+    designs = make_exog_designs(exog, T)               # full-series designs
+    xs, ps = _designs_to_gpu(designs, device)
+    KX = xs[0].shape[1]
+    fitted = torch.empty((T, G), dtype=torch.float32, device=device)
+    params = torch.empty((G, 1 + KX + 8), dtype=torch.float32,
+                         device=device)
+    fstatus = torch.empty((G,), dtype=torch.uint8, device=device)
+    _C.groupfit_final(yT, xs[0], xs[1], xs[2], ps[0], ps[1], ps[2],
+                      best_order, fitted, params, fstatus)
+    return {
+        "best_order": best_order, "mse": mse, "eval_status": status,
+        "fitted": fitted.t().contiguous(), "params": params,
+        "status": fstatus,
+    }
 
 
 def fitted_values_reference(y: np.ndarray, exog: np.ndarray,

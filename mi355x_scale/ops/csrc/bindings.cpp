@@ -34,7 +34,75 @@ void normalize_u8_to_bf16(torch::Tensor in, torch::Tensor out,
       (long long)in.numel(), stream.stream());
 }
 
+extern "C" void launch_groupfit_eval(
+    const float* yT, const float* xc0, const float* xc1, const float* xc2,
+    const float* pj0, const float* pj1, const float* pj2, const int* orders,
+    float* mse, unsigned char* statusv, int T, long long G, int S, int C,
+    int KX, hipStream_t stream);
+
+extern "C" void launch_groupfit_final(
+    const float* yT, const float* xc0, const float* xc1, const float* xc2,
+    const float* pj0, const float* pj1, const float* pj2,
+    const int* best_order, float* fitted, float* params,
+    unsigned char* statusv, int T, long long G, int KX, hipStream_t stream);
+
+static void _check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              name, " must be contiguous f32 on GPU");
+}
+
+// yT: [T][G] time-major; xc{d}: [T-d][KX]; pj{d}: [KX][S-d];
+// orders: [C][3] i32; mse: [C][G] f32 out; status: [C][G] u8 out
+void groupfit_eval(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
+                   torch::Tensor xc2, torch::Tensor pj0, torch::Tensor pj1,
+                   torch::Tensor pj2, torch::Tensor orders,
+                   torch::Tensor mse, torch::Tensor status, int64_t S) {
+  _check_f32(yT, "yT"); _check_f32(xc0, "xc0"); _check_f32(xc1, "xc1");
+  _check_f32(xc2, "xc2"); _check_f32(pj0, "pj0"); _check_f32(pj1, "pj1");
+  _check_f32(pj2, "pj2"); _check_f32(mse, "mse");
+  TORCH_CHECK(orders.scalar_type() == torch::kInt32 && orders.is_cuda());
+  TORCH_CHECK(status.scalar_type() == torch::kUInt8 && status.is_cuda());
+  int T = yT.size(0);
+  long long G = yT.size(1);
+  int C = orders.size(0);
+  TORCH_CHECK(S > 8 && S <= T, "need 8 < S <= T");
+  TORCH_CHECK(mse.size(0) == C && mse.size(1) == G);
+  auto stream = at::cuda::getCurrentHIPStream();
+  launch_groupfit_eval(
+      yT.data_ptr<float>(), xc0.data_ptr<float>(), xc1.data_ptr<float>(),
+      xc2.data_ptr<float>(), pj0.data_ptr<float>(), pj1.data_ptr<float>(),
+      pj2.data_ptr<float>(), orders.data_ptr<int>(), mse.data_ptr<float>(),
+      status.data_ptr<uint8_t>(), T, G, (int)S, C, (int)xc0.size(1),
+      stream.stream());
+}
+
+void groupfit_final(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
+                    torch::Tensor xc2, torch::Tensor pj0, torch::Tensor pj1,
+                    torch::Tensor pj2, torch::Tensor best_order,
+                    torch::Tensor fitted, torch::Tensor params,
+                    torch::Tensor status) {
+  _check_f32(yT, "yT"); _check_f32(fitted, "fitted");
+  _check_f32(params, "params");
+  TORCH_CHECK(best_order.scalar_type() == torch::kInt32 &&
+              best_order.is_cuda());
+  int T = yT.size(0);
+  long long G = yT.size(1);
+  TORCH_CHECK(fitted.size(0) == T && fitted.size(1) == G);
+  auto stream = at::cuda::getCurrentHIPStream();
+  launch_groupfit_final(
+      yT.data_ptr<float>(), xc0.data_ptr<float>(), xc1.data_ptr<float>(),
+      xc2.data_ptr<float>(), pj0.data_ptr<float>(), pj1.data_ptr<float>(),
+      pj2.data_ptr<float>(), best_order.data_ptr<int>(),
+      fitted.data_ptr<float>(), params.data_ptr<float>(),
+      status.data_ptr<uint8_t>(), T, G, (int)xc0.size(1), stream.stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
+  m.def("groupfit_eval", &groupfit_eval,
+        "batched per-group ARIMAX candidate evaluation (validation MSE)");
+  m.def("groupfit_final", &groupfit_final,
+        "batched per-group final fit (fitted values + params)");
 }

@@ -28,6 +28,9 @@ class ScaleModule(nn.Module):
 
     def log(self, key: str, value, prog_bar: bool = False,
             sync_dist: bool = False, **_):
+        if (torch.is_tensor(value) and value.is_cuda
+                and torch.cuda.is_current_stream_capturing()):
+            return  # inside hipGraph capture: no host readbacks possible
         v = float(value.detach() if torch.is_tensor(value) else value)
         self._logged[key] = v
         if self.trainer is not None:

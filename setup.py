@@ -15,6 +15,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 SRC = [
     "mi355x_scale/ops/csrc/bindings.cpp",
     "mi355x_scale/ops/csrc/preprocess.hip",
+    "mi355x_scale/ops/csrc/groupfit.hip",
 ]
 
 setup(
