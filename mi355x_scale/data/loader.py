@@ -104,16 +104,20 @@ class DeviceLoader:
     """
 
     def __init__(self, loader: DataLoader, device, depth: int = 2,
-                 dtype_map: Optional[Dict[str, torch.dtype]] = None):
+                 dtype_map: Optional[Dict[str, torch.dtype]] = None,
+                 stagers: int = 2):
         self.loader = loader
         self.device = torch.device(device)
         self.depth = max(2, depth)
         self.dtype_map = dtype_map or {}
+        self.stagers = max(1, stagers)
         self._use_cuda = self.device.type == "cuda"
         if self._use_cuda:
-            self.copy_stream = torch.cuda.Stream(device=self.device)
-        self._pinned: Dict[int, Dict[str, torch.Tensor]] = {}
-        self._slot_events: Dict[int, torch.cuda.Event] = {}
+            # one copy stream per stager thread
+            self.copy_streams = [torch.cuda.Stream(device=self.device)
+                                 for _ in range(self.stagers)]
+        self._pinned: Dict[tuple, Dict[str, torch.Tensor]] = {}
+        self._slot_events: Dict[tuple, torch.cuda.Event] = {}
 
     def _pin_slot(self, slot: int, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
         # A slot may be rewritten only after its previous H2D copy finished
@@ -145,37 +149,43 @@ class DeviceLoader:
         yield from self._iter_cuda()
 
     def _iter_cuda(self):
-        """Batch assembly + pin-copy + async H2D all run on a background
-        thread (the HIP runtime is thread-safe); the consumer thread only
+        """Batch assembly + pin-copy + async H2D run on ``stagers``
+        background threads (the HIP runtime is thread-safe), each with its
+        own pinned-slot ring and copy stream; the consumer thread only
         dequeues ready (event, device_batch) pairs and inserts a stream
-        wait — so host-side staging never blocks the training step."""
+        wait — host-side staging never blocks the training step, and with
+        2+ stagers a slow batch assembly overlaps the next one."""
         import queue as _q
         import threading
 
-        out_q: "_q.Queue" = _q.Queue(maxsize=self.depth)
+        out_q: "_q.Queue" = _q.Queue(maxsize=self.depth * self.stagers)
         stop = threading.Event()
-        device, copy_stream = self.device, self.copy_stream
+        device = self.device
+        src_it = iter(self.loader)
+        src_lock = threading.Lock()
 
         import os
         import time as _time
         debug = os.environ.get("MI355X_LOADER_DEBUG") == "1"
         stats = {"dequeue": [], "pin": [], "h2d_issue": []} if debug else None
+        stats_lock = threading.Lock()
 
-        def _produce():
+        def _produce(sid: int):
             torch.cuda.set_device(device)
+            copy_stream = self.copy_streams[sid]
             try:
                 slot = 0
-                it = iter(self.loader)
                 while True:
                     t0 = _time.perf_counter()
-                    try:
-                        cpu_batch = next(it)
-                    except StopIteration:
-                        return
+                    with src_lock:
+                        try:
+                            cpu_batch = next(src_it)
+                        except StopIteration:
+                            return
                     if stop.is_set():
                         return
                     t1 = _time.perf_counter()
-                    pinned = self._pin_slot(slot, cpu_batch)
+                    pinned = self._pin_slot((sid, slot), cpu_batch)
                     t2 = _time.perf_counter()
                     with torch.cuda.stream(copy_stream):
                         dev = {}
@@ -189,10 +199,11 @@ class DeviceLoader:
                         ev.record(copy_stream)
                     if debug:
                         t3 = _time.perf_counter()
-                        stats["dequeue"].append(t1 - t0)
-                        stats["pin"].append(t2 - t1)
-                        stats["h2d_issue"].append(t3 - t2)
-                    self._slot_events[slot] = ev
+                        with stats_lock:
+                            stats["dequeue"].append(t1 - t0)
+                            stats["pin"].append(t2 - t1)
+                            stats["h2d_issue"].append(t3 - t2)
+                    self._slot_events[(sid, slot)] = ev
                     slot = (slot + 1) % self.depth
                     while not stop.is_set():
                         try:
@@ -201,15 +212,17 @@ class DeviceLoader:
                         except _q.Full:
                             continue
             finally:
-                if debug and stats["dequeue"]:
+                if debug and sid == 0 and stats["dequeue"]:
                     import numpy as _np
                     import sys
-                    for k, v in stats.items():
-                        a = _np.array(v) * 1e3
-                        print(f"[loader-debug] {k}: p50={_np.median(a):.2f} "
-                              f"p95={_np.percentile(a, 95):.2f} "
-                              f"max={a.max():.2f} n={len(a)}",
-                              file=sys.stderr)
+                    with stats_lock:
+                        for k, v in stats.items():
+                            a = _np.array(v) * 1e3
+                            print(f"[loader-debug] {k}: "
+                                  f"p50={_np.median(a):.2f} "
+                                  f"p95={_np.percentile(a, 95):.2f} "
+                                  f"max={a.max():.2f} n={len(a)}",
+                                  file=sys.stderr)
                 while not stop.is_set():
                     try:
                         out_q.put(None, timeout=0.1)
@@ -217,13 +230,20 @@ class DeviceLoader:
                     except _q.Full:
                         continue
 
-        t = threading.Thread(target=_produce, name="h2d-stager", daemon=True)
-        t.start()
+        threads = [threading.Thread(target=_produce, args=(i,),
+                                    name=f"h2d-stager-{i}", daemon=True)
+                   for i in range(self.stagers)]
+        for t in threads:
+            t.start()
+        done_sentinels = 0
         try:
             while True:
                 item = out_q.get()
                 if item is None:
-                    return
+                    done_sentinels += 1
+                    if done_sentinels == self.stagers:
+                        return
+                    continue
                 ev, dev = item
                 torch.cuda.current_stream(self.device).wait_event(ev)
                 yield dev

@@ -68,6 +68,7 @@ class ImageStreamDataModule(DataModule):
         image_hw=(224, 224),
         device: Optional[torch.device] = None,
         prefetch_depth: int = 2,
+        stagers: int = 2,
         val_fraction_shards: bool = True,
     ):
         self.data_dir = data_dir
@@ -82,6 +83,7 @@ class ImageStreamDataModule(DataModule):
             torch.device("cuda", torch.cuda.current_device())
             if torch.cuda.is_available() else torch.device("cpu"))
         self.prefetch_depth = prefetch_depth
+        self.stagers = stagers
         self._manifest: Optional[DatasetManifest] = None
         self._open_loaders = []
 
@@ -108,7 +110,8 @@ class ImageStreamDataModule(DataModule):
             num_epochs=num_epochs,
         )
         loader = DeviceLoader(DataLoader(reader, self.batch_size),
-                              self.device, depth=self.prefetch_depth)
+                              self.device, depth=self.prefetch_depth,
+                              stagers=self.stagers)
         self._open_loaders.append(loader)
         return loader
 
