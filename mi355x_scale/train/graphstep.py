@@ -65,6 +65,9 @@ class GraphedTrainStep:
             for _ in range(max(1, warmup)):
                 _fwd_bwd()
                 if world_size > 1:
+                    # identical to the replay-time step: keep ranks'
+                    # parameters bit-identical through warmup too
+                    self._allreduce_grads()
                     self.flat_grads.div_(world_size)
                 optimizer.step()
         torch.cuda.current_stream(self.device).wait_stream(s)
@@ -88,12 +91,21 @@ class GraphedTrainStep:
         except RuntimeError:
             self.g_opt = None  # non-capturable optimizer: step eagerly
 
+    def _allreduce_grads(self) -> None:
+        if dist.get_backend() == "gloo":
+            # gloo (CPU test path): stage the flat buffer through host
+            h = self.flat_grads.cpu()
+            dist.all_reduce(h)
+            self.flat_grads.copy_(h)
+        else:
+            dist.all_reduce(self.flat_grads)  # RCCL over xGMI
+
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
         for k, v in batch.items():
             self.static_batch[k].copy_(v, non_blocking=True)
         self.g_fwd_bwd.replay()
         if self.world_size > 1:
-            dist.all_reduce(self.flat_grads)
+            self._allreduce_grads()
         if self.g_opt is not None:
             self.g_opt.replay()
         else:
