@@ -79,6 +79,10 @@ def main() -> None:
     args = ap.parse_args()
 
     use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        # MIOpen exhaustive find: shapes are static (fixed batch), so the
+        # one-time search during warmup buys the fastest conv algos.
+        torch.backends.cudnn.benchmark = True
     image_hw = (224, 224)
     batch = args.batch_size
     if not use_cuda:  # CPU smoke config (no GPU in dev container)
