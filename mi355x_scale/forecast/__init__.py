@@ -7,3 +7,4 @@ from .pipeline import (add_exo_variables, split_train_score_data,  # noqa: F401
                        run_fine_grained_forecast,
                        run_fine_grained_forecast_gpu, TUNING_SCHEMA,
                        FORECAST_HORIZON, SEARCH_SPACE)
+from .holtwinters import ExponentialSmoothing, HoltWintersResults  # noqa: F401

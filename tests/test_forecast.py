@@ -139,3 +139,39 @@ def test_run_fine_grained_forecast_small():
     assert out["SKU"].nunique() == 4
     assert len(out) == 4 * 120
     assert np.isfinite(out["Demand_Fitted"]).all()
+
+
+def test_holt_winters_variants():
+    """The four walkthrough variants (ref group_apply/02_...py:143-188):
+    simple, trend, damped trend, trend+seasonal."""
+    from mi355x_scale.forecast import ExponentialSmoothing
+    rng = np.random.default_rng(3)
+    t = np.arange(208.0)
+    season = 10 * np.sin(2 * np.pi * t / 52)
+    y = 100 + 0.5 * t + season + rng.standard_normal(208)
+
+    simple = ExponentialSmoothing(y).fit()
+    trend = ExponentialSmoothing(y, trend="add").fit()
+    damped = ExponentialSmoothing(y, trend="add", damped_trend=True).fit()
+    hw = ExponentialSmoothing(y, trend="add", seasonal="add",
+                              seasonal_periods=52).fit()
+    # one-step fit: seasonal+trend at least matches simple smoothing
+    assert hw.sse < simple.sse
+    assert np.isfinite(trend.sse) and np.isfinite(damped.sse)
+    # the long-horizon forecast is where seasonality pays: the 52-step
+    # dynamic forecast must track trend+sine closely (simple smoothing
+    # would be off by up to ~23 = trend drift + amplitude)
+    fc = hw.forecast(52)
+    assert len(fc) == 52
+    expect = 100 + 0.5 * (t[-1] + np.arange(1, 53)) + \
+        10 * np.sin(2 * np.pi * (t[-1] + np.arange(1, 53)) / 52)
+    assert np.abs(fc - expect).mean() < 3.0
+
+
+def test_holt_winters_multiplicative():
+    from mi355x_scale.forecast import ExponentialSmoothing
+    t = np.arange(156.0)
+    y = (50 + t) * (1 + 0.2 * np.sin(2 * np.pi * t / 52))
+    fit = ExponentialSmoothing(y, trend="add", seasonal="mul",
+                               seasonal_periods=52).fit()
+    assert fit.sse / len(y) < np.var(y) * 0.1
