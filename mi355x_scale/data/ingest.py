@@ -1,0 +1,124 @@
+"""Image-dataset ingest: directory of images → training-ready parquet.
+
+Native equivalent of the reference's ImageNet preparation notebook
+(``deep_learning/1.data-preparation.py``): parallel file copy/decode
+(``:48-74`` 100-thread copy), label extraction from annotations
+(``:150-169``), monotonic id assignment (``:181-186``) and Delta writes
+(``:193-205``) become: a thread-pool PIL decode → resize/center-crop →
+fixed-size-binary parquet rows + a label-vocabulary JSON (the
+``object_id → contiguous index`` dict of ``deep_learning/2...py:116-125``).
+
+The output is directly consumable by ``ImageStreamDataModule`` — images
+are stored at training resolution so the GPU-side fused normalize kernel
+is the only remaining transform.
+"""
+from __future__ import annotations
+
+import json
+import os
+from concurrent.futures import ThreadPoolExecutor
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+
+IMG_EXTS = (".jpg", ".jpeg", ".png", ".bmp", ".webp")
+
+
+def discover_images(src_dir: str) -> List[Tuple[str, str]]:
+    """[(path, label)] — label is the immediate parent directory name
+    (ImageNet-style layout)."""
+    out = []
+    for dirpath, _, files in os.walk(src_dir):
+        label = os.path.basename(dirpath)
+        for f in sorted(files):
+            if f.lower().endswith(IMG_EXTS):
+                out.append((os.path.join(dirpath, f), label))
+    return out
+
+
+def build_label_vocab(items: List[Tuple[str, str]]) -> Dict[str, int]:
+    """Distinct labels → contiguous class indices (the reference's
+    driver-side vocabulary, ``deep_learning/2...py:116-125``)."""
+    return {lbl: i for i, lbl in
+            enumerate(sorted({lbl for _, lbl in items}))}
+
+
+def _decode_resize(path: str, hw: Tuple[int, int]) -> Optional[bytes]:
+    """PIL decode → resize(short side) → center-crop — the reference's
+    CPU transform (``deep_learning/2...py:282-296``) applied ONCE at
+    ingest instead of on every epoch."""
+    from PIL import Image
+    h, w = hw
+    try:
+        with Image.open(path) as im:
+            im = im.convert("RGB")
+            scale = max(h / im.height, w / im.width) * 256.0 / 224.0
+            nh, nw = int(round(im.height * scale)), int(round(im.width * scale))
+            im = im.resize((nw, nh))
+            left = (nw - w) // 2
+            top = (nh - h) // 2
+            im = im.crop((left, top, left + w, top + h))
+            return np.asarray(im, dtype=np.uint8).tobytes()
+    except Exception:
+        return None
+
+
+def ingest_image_directory(
+    src_dir: str,
+    out_dir: str,
+    image_hw: Tuple[int, int] = (224, 224),
+    rows_per_group: int = 212,
+    rows_per_file: int = 2120,
+    workers: int = 32,
+) -> Dict:
+    """Decode every image under ``src_dir`` on a thread pool and write the
+    parquet dataset + ``label_vocab.json``. Returns a manifest summary
+    (files, rows, skipped)."""
+    os.makedirs(out_dir, exist_ok=True)
+    items = discover_images(src_dir)
+    if not items:
+        raise FileNotFoundError(f"no images under {src_dir}")
+    vocab = build_label_vocab(items)
+    h, w = image_hw
+    row_bytes = h * w * 3
+
+    skipped = 0
+    buf_img: List[bytes] = []
+    buf_lbl: List[int] = []
+    file_idx = 0
+    total = 0
+
+    def flush():
+        nonlocal file_idx, total
+        if not buf_img:
+            return
+        table = pa.table({
+            "image": pa.array(buf_img, type=pa.binary(row_bytes)),
+            "label": pa.array(buf_lbl, type=pa.int64()),
+        })
+        path = os.path.join(out_dir, f"part-{file_idx:05d}.parquet")
+        pq.write_table(table, path, row_group_size=rows_per_group,
+                       compression="none")
+        file_idx += 1
+        total += len(buf_img)
+        buf_img.clear()
+        buf_lbl.clear()
+
+    with ThreadPoolExecutor(max_workers=workers) as pool:
+        for (path, label), blob in zip(
+                items, pool.map(lambda it: _decode_resize(it[0], image_hw),
+                                items, chunksize=16)):
+            if blob is None:
+                skipped += 1
+                continue
+            buf_img.append(blob)
+            buf_lbl.append(vocab[label])
+            if len(buf_img) >= rows_per_file:
+                flush()
+    flush()
+    with open(os.path.join(out_dir, "label_vocab.json"), "w") as f:
+        json.dump(vocab, f, indent=0, sort_keys=True)
+    return {"rows": total, "files": file_idx, "skipped": skipped,
+            "num_classes": len(vocab), "out_dir": out_dir}
