@@ -55,7 +55,8 @@ class TorchDistributor:
     1-process degenerate case)."""
 
     def __init__(self, num_processes: int = 1, local_mode: bool = True,
-                 use_gpu: bool = True, env: dict = None):
+                 use_gpu: bool = True, env: dict = None,
+                 max_restarts: int = 0):
         if not local_mode:
             raise NotImplementedError(
                 "multi-node launch is out of scope on a single MI355X node; "
@@ -64,8 +65,27 @@ class TorchDistributor:
         self.num_processes = int(num_processes)
         self.use_gpu = use_gpu
         self.env = env or {}
+        # Rank-failure handling: a crashed rank fails the
+        # gang; with max_restarts > 0 the whole gang is relaunched (the
+        # Spark-task-retry equivalent — state recovery is the training
+        # loop's checkpoint/resume job).
+        self.max_restarts = int(max_restarts)
 
     def run(self, fn: Callable, *args) -> Any:
+        last_err: Exception = None
+        for attempt in range(self.max_restarts + 1):
+            try:
+                return self._run_once(fn, *args)
+            except RuntimeError as e:
+                last_err = e
+                if attempt < self.max_restarts:
+                    import sys
+                    print(f"[TorchDistributor] gang failed "
+                          f"(attempt {attempt + 1}); relaunching: {e}",
+                          file=sys.stderr)
+        raise last_err
+
+    def _run_once(self, fn: Callable, *args) -> Any:
         if self.num_processes <= 1:
             # Degenerate path, like the reference's direct call
             # (deep_learning/2...py:425-428): run in-process, world size 1.

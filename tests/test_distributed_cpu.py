@@ -72,3 +72,39 @@ def test_shm_broadcast_roundtrip():
     assert np.array_equal(bc2.value, data)
     bc.unpersist()
     assert not os.path.exists(bc.path)
+
+
+_fail_marker = None
+
+
+def _flaky_worker(marker_path):
+    import os
+    # fail on the first gang attempt, succeed on the second
+    if not os.path.exists(marker_path):
+        if int(os.environ["RANK"]) == 1:
+            open(marker_path, "w").write("1")
+            raise RuntimeError("injected rank failure")
+        import time
+        time.sleep(1.0)
+        raise RuntimeError("gang abort")
+    return "recovered"
+
+
+def test_launcher_relaunch_after_rank_failure(tmp_path):
+    """rank failure -> gang relaunch (max_restarts)."""
+    marker = str(tmp_path / "failed_once")
+    out = TorchDistributor(num_processes=2, use_gpu=False,
+                           max_restarts=1).run(_flaky_worker, marker)
+    assert out == "recovered"
+
+
+def _always_fail(m):
+    raise RuntimeError("always")
+
+
+def test_launcher_no_restart_raises(tmp_path):
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        TorchDistributor(num_processes=2, use_gpu=False,
+                         max_restarts=1).run(_always_fail,
+                                             str(tmp_path / "nope"))
