@@ -52,6 +52,7 @@ class Trainer:
         precision: str = "bf16-mixed",
         bucket_cap_mb: int = 32,
         log_every_n_steps: int = 50,
+        use_hipgraph: str = "auto",
     ):
         self.strategy = strategy
         self.max_epochs = max_epochs
@@ -66,6 +67,10 @@ class Trainer:
         self.precision = precision
         self.bucket_cap_mb = bucket_cap_mb
         self.log_every_n_steps = log_every_n_steps
+        # "auto": capture the train step as a hipGraph on GPU (single
+        # replay per step; DP becomes one flat RCCL all-reduce between
+        # replays). "off"/False disables (eager DDP path).
+        self.use_hipgraph = use_hipgraph
         self.ctx: Optional[DistContext] = None
         self.checkpoint_callback = None
         self._sync_accum = {}
@@ -99,9 +104,11 @@ class Trainer:
         if getattr(model, "channels_last", False) and device.type == "cuda":
             model.to(memory_format=torch.channels_last)
 
-        use_ddp = (self.strategy == "ddp") or (
+        graph_on = (device.type == "cuda"
+                    and self.use_hipgraph in ("auto", "on", True))
+        use_ddp = not graph_on and ((self.strategy == "ddp") or (
             self.strategy == "auto" and ctx.world_size > 1
-        )
+        ))
         wrapped = None
         if use_ddp and ctx.world_size > 1:
             kwargs = dict(
@@ -124,6 +131,7 @@ class Trainer:
                      if self.precision.startswith("bf16") else None)
         autocast_on = amp_dtype is not None and device.type == "cuda"
 
+        graphed = None
         step = 0
         for epoch in range(self.max_epochs):
             model.current_epoch = epoch
@@ -139,15 +147,27 @@ class Trainer:
                 except StopIteration:
                     break
                 batch = _move(batch, device)
-                with torch.autocast(device_type="cuda", dtype=amp_dtype,
-                                    enabled=autocast_on):
-                    if wrapped is not None:
-                        loss = wrapped(batch, n_batches)
-                    else:
-                        loss = model.training_step(batch, n_batches)
-                optimizer.zero_grad(set_to_none=True)
-                loss.backward()
-                optimizer.step()
+                if graph_on and graphed is None:
+                    from .graphstep import GraphedTrainStep
+                    graphed = GraphedTrainStep(
+                        model, optimizer, batch,
+                        autocast_dtype=amp_dtype if autocast_on else None,
+                        world_size=ctx.world_size)
+                if graphed is not None:
+                    loss = graphed.step(batch)
+                    if (step + 1) % self.log_every_n_steps == 0:
+                        model.log("train_loss", float(loss.item()))
+                else:
+                    with torch.autocast(device_type="cuda",
+                                        dtype=amp_dtype,
+                                        enabled=autocast_on):
+                        if wrapped is not None:
+                            loss = wrapped(batch, n_batches)
+                        else:
+                            loss = model.training_step(batch, n_batches)
+                    optimizer.zero_grad(set_to_none=True)
+                    loss.backward()
+                    optimizer.step()
                 n_batches += 1
                 step += 1
                 model.global_step = step

@@ -73,3 +73,22 @@ def test_graphed_step_loss_finite_and_changing():
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]  # optimizing the same batch must descend
+
+
+@pytest.mark.gpu
+def test_trainer_hipgraph_mode(tmp_path):
+    """Trainer with use_hipgraph='auto' on GPU: trains, checkpoints."""
+    import os
+    from mi355x_scale.data.generator import write_image_parquet
+    from mi355x_scale.train import ImageStreamDataModule, Trainer
+    d = str(tmp_path / "imgs")
+    write_image_parquet(d, num_rows=64, image_hw=(64, 64),
+                        rows_per_group=16, rows_per_file=32)
+    model = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
+    dm = ImageStreamDataModule(d, batch_size=16, workers_count=2,
+                               image_hw=(64, 64))
+    trainer = Trainer(max_epochs=1, limit_train_batches=4,
+                      limit_val_batches=2,
+                      default_root_dir=str(tmp_path / "ckpt"))
+    trainer.fit(model, dm)
+    assert os.path.exists(tmp_path / "ckpt" / "last.ckpt")
