@@ -175,3 +175,23 @@ def test_holt_winters_multiplicative():
     fit = ExponentialSmoothing(y, trend="add", seasonal="mul",
                                seasonal_periods=52).fit()
     assert fit.sse / len(y) < np.var(y) * 0.1
+
+
+def test_panel_gather_roundtrip():
+    """Vectorized scatter-gather matches the pandas pivot reference."""
+    from mi355x_scale.groupby.gather import panel_from_long, long_from_panel
+    df = generate_demand_data(n_products=2, skus_per_product=3, n_weeks=30)
+    df_shuffled = df.sample(frac=1.0, random_state=0)  # any row order
+    panel, gindex, tvals = panel_from_long(
+        df_shuffled, ["Product", "SKU"], "Date", "Demand")
+    piv = df.pivot_table(index=["Product", "SKU"], columns="Date",
+                         values="Demand", sort=True)
+    assert panel.shape == piv.shape
+    assert np.allclose(panel, piv.to_numpy(), atol=1e-4)
+    back = long_from_panel(panel, gindex, tvals, ["Product", "SKU"],
+                           "Date", [("Demand", panel)])
+    assert len(back) == panel.size
+    merged = back.merge(df, on=["Product", "SKU", "Date"],
+                        suffixes=("_got", "_want"))
+    assert np.allclose(merged["Demand_got"], merged["Demand_want"],
+                       atol=1e-4)
