@@ -56,8 +56,10 @@ class BatchReader:
         seed: Optional[int] = None,
         arrow_transform=None,
     ):
-        if reader_pool_type not in ("thread", "dummy"):
-            raise ValueError(f"reader_pool_type must be 'thread' or 'dummy', got {reader_pool_type!r}")
+        if reader_pool_type not in ("thread", "process", "dummy"):
+            raise ValueError(
+                f"reader_pool_type must be 'thread', 'process' or 'dummy', "
+                f"got {reader_pool_type!r}")
         if (cur_shard is None) != (shard_count is None):
             raise ValueError("cur_shard and shard_count must be given together")
         self.manifest = manifest
@@ -93,11 +95,23 @@ class BatchReader:
         self._pf_cache: Dict[str, pq.ParquetFile] = {}
         self._decode_times: List[float] = []
         self._pf_lock = threading.Lock()
+        self._proc_pool = None
         if reader_pool_type == "thread":
             for i in range(self.workers_count):
                 t = threading.Thread(target=self._worker, name=f"rg-reader-{i}", daemon=True)
                 t.start()
                 self._threads.append(t)
+        elif reader_pool_type == "process":
+            if transform_spec is not None:
+                raise ValueError(
+                    "process pool supports arrow_transform (picklable) "
+                    "only; use reader_pool_type='thread' for "
+                    "transform_spec")
+            from .process_pool import ProcessReaderPool
+            self._proc_pool = ProcessReaderPool(
+                self._refs, self.workers_count, num_epochs,
+                schema_fields, arrow_transform,
+                results_queue_size)
 
     # -- work distribution ------------------------------------------------
     def _next_ref(self) -> Optional[RowGroupRef]:
@@ -197,6 +211,12 @@ class BatchReader:
         return self
 
     def __next__(self) -> Dict[str, np.ndarray]:
+        if self._proc_pool is not None:
+            try:
+                return next(self._proc_pool)
+            except StopIteration:
+                self.last_row_consumed = True
+                raise
         if self.reader_pool_type == "dummy":
             ref = self._next_ref()
             if ref is None:
@@ -224,6 +244,8 @@ class BatchReader:
         self._threads = []
 
     def close(self) -> None:
+        if self._proc_pool is not None:
+            self._proc_pool.close()
         self.stop()
         self.join()
         if self._decode_times:

@@ -112,3 +112,27 @@ def test_empty_shard_raises(image_parquet):
     man = DatasetManifest.discover(image_parquet)
     with pytest.raises(ValueError):
         BatchReader(man, cur_shard=7, shard_count=8)  # only 6 row groups
+
+
+def test_reader_process_pool(image_parquet):
+    """reader_pool_type='process' (petastorm's third pool type): same
+    row-exactness contract as the thread pool."""
+    r = make_batch_reader(image_parquet, reader_pool_type="process",
+                          workers_count=2, num_epochs=1,
+                          results_queue_size=4)
+    with r:
+        total = sum(len(b["label"]) for b in r)
+    assert total == 96
+
+
+def test_reader_process_pool_infinite_close(image_parquet):
+    r = make_batch_reader(image_parquet, reader_pool_type="process",
+                          workers_count=2, num_epochs=None,
+                          results_queue_size=2)
+    got = 0
+    with r:
+        for b in r:
+            got += len(b["label"])
+            if got > 96:
+                break
+    assert got > 96
