@@ -53,6 +53,7 @@ class Trainer:
         bucket_cap_mb: int = 32,
         log_every_n_steps: int = 50,
         use_hipgraph: str = "auto",
+        resume_from: str = None,
     ):
         self.strategy = strategy
         self.max_epochs = max_epochs
@@ -71,6 +72,7 @@ class Trainer:
         # replay per step; DP becomes one flat RCCL all-reduce between
         # replays). "off"/False disables (eager DDP path).
         self.use_hipgraph = use_hipgraph
+        self.resume_from = resume_from
         self.ctx: Optional[DistContext] = None
         self.checkpoint_callback = None
         self._sync_accum = {}
@@ -122,6 +124,13 @@ class Trainer:
             wrapped = DDP(_TrainStepShim(model), **kwargs)
 
         optimizer = model.configure_optimizers()
+        start_epoch = 0
+        if self.resume_from:
+            state = CheckpointManager.load(self.resume_from, model,
+                                           optimizer,
+                                           map_location=device)
+            start_epoch = state["epoch"] + 1
+            model.global_step = state["step"]
         if self.enable_checkpointing:
             self.checkpoint_callback = CheckpointManager(
                 self.default_root_dir, monitor="val_loss", rank=ctx.rank)
@@ -132,8 +141,8 @@ class Trainer:
         autocast_on = amp_dtype is not None and device.type == "cuda"
 
         graphed = None
-        step = 0
-        for epoch in range(self.max_epochs):
+        step = model.global_step
+        for epoch in range(start_epoch, self.max_epochs):
             model.current_epoch = epoch
             (wrapped or model).train()
             loader = datamodule.train_dataloader()

@@ -41,3 +41,25 @@ def test_checkpoint_roundtrip(tmp_path):
     assert state["step"] == 5
     for a, b in zip(model.parameters(), model2.parameters()):
         assert torch.equal(a, b)
+
+
+def test_trainer_resume(image_parquet, tmp_path):
+    """Checkpoint → resume: epoch/step restored, training continues."""
+    import torch
+    model = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
+    dm = ImageStreamDataModule(image_parquet, batch_size=16,
+                               workers_count=1, image_hw=(32, 32),
+                               device=torch.device("cpu"))
+    t1 = Trainer(max_epochs=1, limit_train_batches=2, limit_val_batches=1,
+                 precision="fp32", default_root_dir=str(tmp_path))
+    t1.fit(model, dm)
+
+    model2 = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
+    dm2 = ImageStreamDataModule(image_parquet, batch_size=16,
+                                workers_count=1, image_hw=(32, 32),
+                                device=torch.device("cpu"))
+    t2 = Trainer(max_epochs=2, limit_train_batches=2, limit_val_batches=1,
+                 precision="fp32", default_root_dir=str(tmp_path / "b"),
+                 resume_from=str(tmp_path / "last.ckpt"))
+    t2.fit(model2, dm2)
+    assert model2.global_step > 2  # continued past the restored step
