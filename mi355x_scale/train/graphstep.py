@@ -73,9 +73,13 @@ class GraphedTrainStep:
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
 
-        # Capture forward+backward (one graph)...
+        # Capture forward+backward (one graph). thread_local capture
+        # mode: the streaming loader's stager threads keep issuing H2D
+        # copies on their own streams during capture — global mode would
+        # abort the process on their first HIP call.
         self.g_fwd_bwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwd_bwd):
+        with torch.cuda.graph(self.g_fwd_bwd,
+                              capture_error_mode="thread_local"):
             self.static_loss = _fwd_bwd()
 
         # ...and the optimizer step (second graph; the eager RCCL
@@ -83,7 +87,7 @@ class GraphedTrainStep:
         self.g_opt = None
         try:
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 if world_size > 1:
                     self.flat_grads.div_(world_size)
                 optimizer.step()
