@@ -221,28 +221,63 @@ def _designs_to_gpu(designs: List[ExogDesign], device):
     return xs, ps
 
 
-def batched_eval_gpu(y, exog, orders, train_len: int, device="cuda"):
+def _to_yT(y, device):
+    import torch
+    y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32)
+    return y_t.t().contiguous().to(device)
+
+
+def _mfma_projection(yT, ps, train_len: int, device):
+    """Stage-1 via the MFMA design-matrix GEMM for each d: Wc/wm from the
+    diff+center kernel, beta = P_d @ Wc on the f32 matrix cores."""
+    import torch
+    from ..ops import _C
+    T, G = yT.shape
+    betas, wms = [], []
+    for d in range(3):
+        n = train_len - d
+        wc = torch.empty((n, G), dtype=torch.float32, device=device)
+        wm = torch.empty((G,), dtype=torch.float32, device=device)
+        _C.diff_center(yT, wc, wm, train_len, d)
+        beta = torch.empty((ps[d].shape[0], G), dtype=torch.float32,
+                           device=device)
+        _C.exog_project_mfma(ps[d], wc, beta)
+        betas.append(beta)
+        wms.append(wm)
+    return betas, wms
+
+
+def batched_eval_gpu(y, exog, orders, train_len: int, device="cuda",
+                     yT=None, use_mfma: bool = True):
     """GPU candidate evaluation: returns (mse [G,C] torch.f32, status
     [G,C] torch.u8). ``y`` is [G,T] (numpy or torch)."""
     import torch
     from ..ops import require_ext, _C
     require_ext()
-    y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32)
-    G, T = y_t.shape
-    yT = y_t.t().contiguous().to(device)
+    if yT is None:
+        yT = _to_yT(y, device)
+    T, G = yT.shape
     designs = make_exog_designs(exog, train_len)
     xs, ps = _designs_to_gpu(designs, device)
+    if use_mfma:
+        betas, wms = _mfma_projection(yT, ps, train_len, device)
+    else:
+        empty = torch.empty(0, dtype=torch.float32, device=device)
+        betas, wms = [empty] * 3, [empty] * 3
     orders_t = torch.tensor(list(orders), dtype=torch.int32,
                             device=device).reshape(-1, 3).contiguous()
     C = orders_t.shape[0]
     mse = torch.empty((C, G), dtype=torch.float32, device=device)
     status = torch.empty((C, G), dtype=torch.uint8, device=device)
     _C.groupfit_eval(yT, xs[0], xs[1], xs[2], ps[0], ps[1], ps[2],
-                     orders_t, mse, status, train_len)
+                     orders_t, mse, status, train_len,
+                     betas[0], betas[1], betas[2],
+                     wms[0], wms[1], wms[2])
     return mse.t().contiguous(), status.t().contiguous()
 
 
-def batched_fit_gpu(y, exog, orders, train_len: int, device="cuda"):
+def batched_fit_gpu(y, exog, orders, train_len: int, device="cuda",
+                    use_mfma: bool = True):
     """Full W1 GPU pipeline: evaluate candidates, pick the best per group,
     final-fit on the whole series. Returns a dict with ``best_order``
     [G,3], ``mse`` [G,C], ``fitted`` [G,T], ``params`` [G,1+KX+8],
@@ -250,10 +285,10 @@ def batched_fit_gpu(y, exog, orders, train_len: int, device="cuda"):
     import torch
     from ..ops import require_ext, _C
     require_ext()
-    y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32)
-    G, T = y_t.shape
-    yT = y_t.t().contiguous().to(device)
-    mse, status = batched_eval_gpu(y, exog, orders, train_len, device)
+    yT = _to_yT(y, device)
+    T, G = yT.shape
+    mse, status = batched_eval_gpu(y, exog, orders, train_len, device,
+                                   yT=yT, use_mfma=use_mfma)
     orders_t = torch.tensor(list(orders), dtype=torch.int32, device=device
                             ).reshape(-1, 3)
     best_ci = mse.argmin(dim=1)                        # [G]

@@ -37,8 +37,19 @@ void normalize_u8_to_bf16(torch::Tensor in, torch::Tensor out,
 extern "C" void launch_groupfit_eval(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
     const float* pj0, const float* pj1, const float* pj2, const int* orders,
-    float* mse, unsigned char* statusv, int T, long long G, int S, int C,
+    float* mse, unsigned char* statusv,
+    const float* b0, const float* b1, const float* b2,
+    const float* w0, const float* w1, const float* w2,
+    int T, long long G, int S, int C,
     int KX, hipStream_t stream);
+
+extern "C" void launch_diff_center(const float* yT, float* wc, float* wm,
+                                   int T, long long G, int S, int d,
+                                   hipStream_t stream);
+
+extern "C" void launch_exog_project_mfma(const float* P, const float* wc,
+                                         float* beta, int n, long long G,
+                                         int KX, hipStream_t stream);
 
 extern "C" void launch_groupfit_final(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
@@ -57,7 +68,9 @@ static void _check_f32(const torch::Tensor& t, const char* name) {
 void groupfit_eval(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
                    torch::Tensor xc2, torch::Tensor pj0, torch::Tensor pj1,
                    torch::Tensor pj2, torch::Tensor orders,
-                   torch::Tensor mse, torch::Tensor status, int64_t S) {
+                   torch::Tensor mse, torch::Tensor status, int64_t S,
+                   torch::Tensor b0, torch::Tensor b1, torch::Tensor b2,
+                   torch::Tensor w0, torch::Tensor w1, torch::Tensor w2) {
   _check_f32(yT, "yT"); _check_f32(xc0, "xc0"); _check_f32(xc1, "xc1");
   _check_f32(xc2, "xc2"); _check_f32(pj0, "pj0"); _check_f32(pj1, "pj1");
   _check_f32(pj2, "pj2"); _check_f32(mse, "mse");
@@ -69,11 +82,16 @@ void groupfit_eval(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
   TORCH_CHECK(S > 8 && S <= T, "need 8 < S <= T");
   TORCH_CHECK(mse.size(0) == C && mse.size(1) == G);
   auto stream = at::cuda::getCurrentHIPStream();
+  auto fp = [](torch::Tensor& t) -> const float* {
+    return t.numel() ? t.data_ptr<float>() : nullptr;
+  };
   launch_groupfit_eval(
       yT.data_ptr<float>(), xc0.data_ptr<float>(), xc1.data_ptr<float>(),
       xc2.data_ptr<float>(), pj0.data_ptr<float>(), pj1.data_ptr<float>(),
       pj2.data_ptr<float>(), orders.data_ptr<int>(), mse.data_ptr<float>(),
-      status.data_ptr<uint8_t>(), T, G, (int)S, C, (int)xc0.size(1),
+      status.data_ptr<uint8_t>(),
+      fp(b0), fp(b1), fp(b2), fp(w0), fp(w1), fp(w2),
+      T, G, (int)S, C, (int)xc0.size(1),
       stream.stream());
 }
 
@@ -98,6 +116,30 @@ void groupfit_final(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
       status.data_ptr<uint8_t>(), T, G, (int)xc0.size(1), stream.stream());
 }
 
+// Wc [n][G], wm [G] from the time-major panel (differenced d times,
+// centered with the per-group train mean).
+void diff_center(torch::Tensor yT, torch::Tensor wc, torch::Tensor wm,
+                 int64_t S, int64_t d) {
+  _check_f32(yT, "yT"); _check_f32(wc, "wc"); _check_f32(wm, "wm");
+  auto stream = at::cuda::getCurrentHIPStream();
+  launch_diff_center(yT.data_ptr<float>(), wc.data_ptr<float>(),
+                     wm.data_ptr<float>(), yT.size(0), yT.size(1),
+                     (int)S, (int)d, stream.stream());
+}
+
+// beta[KX][G] = P[KX][n] @ wc[n][G] on the f32 matrix cores.
+void exog_project_mfma(torch::Tensor P, torch::Tensor wc,
+                       torch::Tensor beta) {
+  _check_f32(P, "P"); _check_f32(wc, "wc"); _check_f32(beta, "beta");
+  TORCH_CHECK(P.size(1) == wc.size(0), "K mismatch");
+  TORCH_CHECK(beta.size(0) == P.size(0) && beta.size(1) == wc.size(1));
+  TORCH_CHECK(P.size(0) <= 16, "KX must fit one MFMA row tile");
+  auto stream = at::cuda::getCurrentHIPStream();
+  launch_exog_project_mfma(P.data_ptr<float>(), wc.data_ptr<float>(),
+                           beta.data_ptr<float>(), P.size(1), wc.size(1),
+                           P.size(0), stream.stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
@@ -105,4 +147,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched per-group ARIMAX candidate evaluation (validation MSE)");
   m.def("groupfit_final", &groupfit_final,
         "batched per-group final fit (fitted values + params)");
+  m.def("diff_center", &diff_center,
+        "difference + per-group centering of the time-major panel");
+  m.def("exog_project_mfma", &exog_project_mfma,
+        "design-matrix GEMM beta = P @ Wc on f32 MFMA");
 }

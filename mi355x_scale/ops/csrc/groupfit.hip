@@ -114,7 +114,9 @@ __device__ __forceinline__ float fit_lane(
     const float* __restrict__ yT, const float* __restrict__ xc,
     const float* __restrict__ PJ, float* U, float* E, float* st,
     int lane, long long g, long long G, int S, int KX, int p, int d, int q,
-    bool* ok_out) {
+    bool* ok_out,
+    const float* __restrict__ beta_pre,  // [KX][G] from the MFMA
+    const float* __restrict__ wm_pre) {  // [G]     projection, or null
   const int n = S - d;
   // ---- stage 0: difference + mean (stream y, write w into U)
   float y0 = 0.f, y1 = 0.f;
@@ -131,16 +133,22 @@ __device__ __forceinline__ float fit_lane(
       wsum += w;
     }
   }
-  const float wm = wsum / (float)n;
-  // ---- stage 1: beta = P @ (w - wm); u = wc - Xc beta
+  const float wm = wm_pre ? wm_pre[g] : (wsum / (float)n);
+  // ---- stage 1: beta = P @ (w - wm); u = wc - Xc beta. beta comes from
+  // the MFMA design-matrix GEMM (ops/csrc/mfma_project.hip) when
+  // precomputed, else from the per-lane matvec.
 #pragma unroll
   for (int k = 0; k < MAXK; ++k) BETA(k) = 0.0f;
-  for (int k = 0; k < KX; ++k) {
-    float acc = 0.0f;
-    const float* Pk = PJ + (long long)k * n;
-    for (int t = 0; t < n; ++t)
-      acc += Pk[t] * (U[t * WAVE + lane] - wm);
-    BETA(k) = acc;
+  if (beta_pre) {
+    for (int k = 0; k < KX; ++k) BETA(k) = beta_pre[(long long)k * G + g];
+  } else {
+    for (int k = 0; k < KX; ++k) {
+      float acc = 0.0f;
+      const float* Pk = PJ + (long long)k * n;
+      for (int t = 0; t < n; ++t)
+        acc += Pk[t] * (U[t * WAVE + lane] - wm);
+      BETA(k) = acc;
+    }
   }
   for (int t = 0; t < n; ++t) {
     float reg = 0.0f;
@@ -252,6 +260,10 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_eval_kernel(
     const int* __restrict__ orders,       // [C][3]
     float* __restrict__ mse,              // [C][G]
     unsigned char* __restrict__ statusv,  // [C][G]
+    const float* __restrict__ b0, const float* __restrict__ b1,
+    const float* __restrict__ b2,         // [KX][G] per d, or null
+    const float* __restrict__ w0, const float* __restrict__ w1,
+    const float* __restrict__ w2,         // [G] per d, or null
     int T, long long G, int S, int C, int KX) {
   extern __shared__ float lds[];
   float* U = lds;
@@ -271,9 +283,11 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_eval_kernel(
     const float* PJ = (d == 0) ? pj0 : (d == 1) ? pj1 : pj2;
     const int n = S - d;
 
+    const float* bpre = (d == 0) ? b0 : (d == 1) ? b1 : b2;
+    const float* wpre = (d == 0) ? w0 : (d == 1) ? w1 : w2;
     bool ok;
     const float wm = fit_lane(yT, xc, PJ, U, E, st, lane, gg, G, S, KX,
-                              p, d, q, &ok);
+                              p, d, q, &ok, bpre, wpre);
 
     // ---- stage 5: validation forecast + MSE
     float u0 = U[(n - 1) * WAVE + lane];
@@ -349,7 +363,7 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_final_kernel(
 
   bool ok;
   const float wm = fit_lane(yT, xc, PJ, U, E, st, lane, gg, G, S, KX,
-                            p, d, q, &ok);
+                            p, d, q, &ok, nullptr, nullptr);
 
   if (active) {
     const float ph0 = PHI(0), ph1 = PHI(1), ph2 = PHI(2), ph3 = PHI(3);
@@ -404,7 +418,10 @@ static void _set_lds_limit(const void* func, size_t shmem) {
 extern "C" void launch_groupfit_eval(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
     const float* pj0, const float* pj1, const float* pj2, const int* orders,
-    float* mse, unsigned char* statusv, int T, long long G, int S, int C,
+    float* mse, unsigned char* statusv,
+    const float* b0, const float* b1, const float* b2,
+    const float* w0, const float* w1, const float* w2,
+    int T, long long G, int S, int C,
     int KX, hipStream_t stream) {
   const int blocks = (int)((G + WAVE - 1) / WAVE);
   const size_t shmem =
@@ -412,7 +429,7 @@ extern "C" void launch_groupfit_eval(
   _set_lds_limit(reinterpret_cast<const void*>(groupfit_eval_kernel), shmem);
   hipLaunchKernelGGL(groupfit_eval_kernel, dim3(blocks), dim3(WAVE), shmem,
                      stream, yT, xc0, xc1, xc2, pj0, pj1, pj2, orders, mse,
-                     statusv, T, G, S, C, KX);
+                     statusv, b0, b1, b2, w0, w1, w2, T, G, S, C, KX);
 }
 
 extern "C" void launch_groupfit_final(

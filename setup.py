@@ -16,6 +16,7 @@ SRC = [
     "mi355x_scale/ops/csrc/bindings.cpp",
     "mi355x_scale/ops/csrc/preprocess.hip",
     "mi355x_scale/ops/csrc/groupfit.hip",
+    "mi355x_scale/ops/csrc/mfma_project.hip",
 ]
 
 setup(

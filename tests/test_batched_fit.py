@@ -111,3 +111,34 @@ def test_gpu_forecast_pipeline_end_to_end():
     assert np.isfinite(f).all()
     mse = np.mean((d[10:] - f[10:]) ** 2)
     assert mse < 0.6 * np.var(d[10:])
+
+
+@pytest.mark.gpu
+def test_mfma_projection_matches_matmul():
+    """The design-matrix GEMM on f32 MFMA vs torch matmul (exact-f32
+    MFMA: bitwise-class agreement expected, tolerance covers ordering)."""
+    import torch
+    from mi355x_scale.ops import _C
+    rng = np.random.default_rng(0)
+    n, G, KX = 117, 1000, 3
+    P = torch.tensor(rng.standard_normal((KX, n)), dtype=torch.float32,
+                     device="cuda").contiguous()
+    wc = torch.tensor(rng.standard_normal((n, G)), dtype=torch.float32,
+                      device="cuda").contiguous()
+    beta = torch.empty((KX, G), dtype=torch.float32, device="cuda")
+    _C.exog_project_mfma(P, wc, beta)
+    ref = P @ wc
+    assert torch.allclose(beta, ref, atol=1e-4, rtol=1e-5), \
+        (beta - ref).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_mfma_path_equals_lane_path():
+    """Eval MSEs with MFMA-precomputed stage 1 vs the in-kernel matvec."""
+    from mi355x_scale.forecast.batched import batched_eval_gpu
+    y, exog = _dataset(2, 8)
+    m1, s1 = batched_eval_gpu(y, exog, ORDERS, 117, use_mfma=True)
+    m2, s2 = batched_eval_gpu(y, exog, ORDERS, 117, use_mfma=False)
+    import torch
+    rel = ((m1 - m2).abs() / (m2.abs() + 1e-3)).max().item()
+    assert rel < 1e-3, rel
