@@ -84,16 +84,20 @@ class GraphedTrainStep:
 
         # ...and the optimizer step (second graph; the eager RCCL
         # all-reduce slots between the two replays when world_size > 1).
+        # Only optimizers declaring capturable=True are captured — an
+        # exception thrown mid-capture (e.g. fused Adam's capturable
+        # check) leaves the HIP stream in a broken capture state that
+        # later faults, so no try-capture-and-fallback.
         self.g_opt = None
-        try:
+        capturable = all(group.get("capturable", False)
+                         for group in optimizer.param_groups)
+        if capturable:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 if world_size > 1:
                     self.flat_grads.div_(world_size)
                 optimizer.step()
             self.g_opt = g
-        except RuntimeError:
-            self.g_opt = None  # non-capturable optimizer: step eagerly
 
     def _allreduce_grads(self) -> None:
         if dist.get_backend() == "gloo":

@@ -103,12 +103,10 @@ class ImageClassifier(ScaleModule):
 
     def configure_optimizers(self):
         if next(self.parameters()).is_cuda:
-            try:
-                # single fused kernel per step instead of a foreach chain
-                return torch.optim.Adam(self.parameters(), lr=self.lr,
-                                        fused=True)
-            except (RuntimeError, ValueError):
-                pass
+            # capturable: the optimizer step can be recorded into the
+            # hipGraph train step (Trainer use_hipgraph / GraphedTrainStep)
+            return torch.optim.Adam(self.parameters(), lr=self.lr,
+                                    foreach=True, capturable=True)
         return torch.optim.Adam(self.parameters(), lr=self.lr)
 
 
