@@ -18,8 +18,6 @@ MI355X-first choices:
 """
 from __future__ import annotations
 
-import math
-import os
 import time
 from typing import Optional
 
