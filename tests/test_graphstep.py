@@ -82,7 +82,7 @@ def test_trainer_hipgraph_mode(tmp_path):
     from mi355x_scale.data.generator import write_image_parquet
     from mi355x_scale.train import ImageStreamDataModule, Trainer
     d = str(tmp_path / "imgs")
-    write_image_parquet(d, num_rows=64, image_hw=(64, 64),
+    write_image_parquet(d, num_rows=64, image_hw=(64, 64), num_classes=10,
                         rows_per_group=16, rows_per_file=32)
     model = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
     dm = ImageStreamDataModule(d, batch_size=16, workers_count=2,
