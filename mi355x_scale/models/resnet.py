@@ -3,9 +3,12 @@
 The reference trains ``torchvision.models.resnet50``
 (``deep_learning/2.distributed-data-loading-petastorm.py:150``); the
 benchmark model named by BASELINE.json is ResNet-18. torchvision is not a
-dependency here — the architectures are defined directly. Convs/BNs run
-through MIOpen/hipBLASLt via PyTorch-ROCm; channels-last memory format is
-used on GPU (MIOpen's fast path on CDNA).
+dependency here — the architectures are defined directly. Convs run
+through MIOpen via PyTorch-ROCm in channels-last memory format (MIOpen's
+fast conv layout on CDNA); every BatchNorm + residual-add + ReLU group is
+a single ``FusedBNReLU2d`` (hand-written bf16 NHWC HIP kernels,
+ops/csrc/fused_bn.hip) instead of the MIOpen fp32 4-kernel BN path plus
+standalone elementwise passes.
 """
 from __future__ import annotations
 
@@ -13,6 +16,8 @@ from typing import List, Optional, Type, Union
 
 import torch
 import torch.nn as nn
+
+from ..ops.fused_bn import FusedBNReLU2d
 
 
 def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -30,19 +35,15 @@ class BasicBlock(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = _conv3x3(cin, cout, stride)
-        self.bn1 = nn.BatchNorm2d(cout)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = FusedBNReLU2d(cout)
         self.conv2 = _conv3x3(cout, cout)
-        self.bn2 = nn.BatchNorm2d(cout)
+        self.bn2 = FusedBNReLU2d(cout)  # fuses the residual add + relu
         self.downsample = downsample
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        return self.relu(out + identity)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=identity)
 
 
 class Bottleneck(nn.Module):
@@ -52,22 +53,18 @@ class Bottleneck(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = _conv1x1(cin, cout)
-        self.bn1 = nn.BatchNorm2d(cout)
+        self.bn1 = FusedBNReLU2d(cout)
         self.conv2 = _conv3x3(cout, cout, stride)
-        self.bn2 = nn.BatchNorm2d(cout)
+        self.bn2 = FusedBNReLU2d(cout)
         self.conv3 = _conv1x1(cout, cout * self.expansion)
-        self.bn3 = nn.BatchNorm2d(cout * self.expansion)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn3 = FusedBNReLU2d(cout * self.expansion)
         self.downsample = downsample
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        return self.relu(out + identity)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=identity)
 
 
 class ResNet(nn.Module):

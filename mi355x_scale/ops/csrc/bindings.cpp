@@ -140,6 +140,144 @@ void exog_project_mfma(torch::Tensor P, torch::Tensor wc,
                            P.size(0), stream.stream());
 }
 
+// ---------------------------------------------------------------- fused BN
+// NHWC bf16 fused BatchNorm(+residual)(+ReLU); see fused_bn.hip.
+
+extern "C" void launch_bn_fwd_reduce(const void* x, float* accum,
+                                     long long rows, int C,
+                                     hipStream_t stream);
+extern "C" void launch_bn_fwd_finalize(const float* accum, float* mean,
+                                       float* invstd, float* running_mean,
+                                       float* running_var, float momentum,
+                                       float eps, long long rows, int C,
+                                       int update_running, hipStream_t stream);
+extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
+                                    const float* mean, const float* invstd,
+                                    const float* weight, const float* bias,
+                                    long long rows, int C, int relu,
+                                    hipStream_t stream);
+extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
+                                     const void* x, const float* mean,
+                                     const float* invstd, float* accum2,
+                                     long long rows, int C, int relu,
+                                     hipStream_t stream);
+extern "C" void launch_bn_bwd_finalize(const float* accum2,
+                                       const float* invstd,
+                                       const float* weight, float* dweight,
+                                       float* dbias, float* k, long long rows,
+                                       int C, hipStream_t stream);
+extern "C" void launch_bn_bwd_apply(const void* dz, const void* y,
+                                    const void* x, const float* mean,
+                                    const float* invstd, const float* k,
+                                    void* dx, void* dres, long long rows,
+                                    int C, int relu, hipStream_t stream);
+
+// Activations: bf16, NHWC memory order (a channels_last NCHW tensor viewed
+// flat). rows = N*H*W; C must be 8*2^k (ResNet uses 64..512).
+static void _check_bn_act(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16,
+              name, " must be bf16 on GPU");
+}
+
+static long long _bn_rows(const torch::Tensor& x, int64_t C) {
+  TORCH_CHECK(C % 8 == 0 && 256 % (C / 8) == 0,
+              "C must be 8*2^k and <= 2048, got ", C);
+  TORCH_CHECK(x.numel() % C == 0, "numel not divisible by C");
+  return (long long)(x.numel() / C);
+}
+
+void bn_fwd_reduce(torch::Tensor x, torch::Tensor accum, int64_t C) {
+  _check_bn_act(x, "x");
+  _check_f32(accum, "accum");
+  TORCH_CHECK(accum.numel() == 2 * C, "accum must be [2C]");
+  launch_bn_fwd_reduce(x.data_ptr(), accum.data_ptr<float>(),
+                       _bn_rows(x, C), (int)C,
+                       at::cuda::getCurrentHIPStream().stream());
+}
+
+void bn_fwd_finalize(torch::Tensor accum, torch::Tensor mean,
+                     torch::Tensor invstd, torch::Tensor running_mean,
+                     torch::Tensor running_var, double momentum, double eps,
+                     int64_t rows, int64_t C, bool update_running) {
+  _check_f32(accum, "accum"); _check_f32(mean, "mean");
+  _check_f32(invstd, "invstd");
+  _check_f32(running_mean, "running_mean");
+  _check_f32(running_var, "running_var");
+  TORCH_CHECK(C <= 512, "finalize kernel is single-block (C <= 512)");
+  launch_bn_fwd_finalize(accum.data_ptr<float>(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         running_mean.data_ptr<float>(),
+                         running_var.data_ptr<float>(), (float)momentum,
+                         (float)eps, rows, (int)C, update_running ? 1 : 0,
+                         at::cuda::getCurrentHIPStream().stream());
+}
+
+void bn_fwd_apply(torch::Tensor x, torch::Tensor res, torch::Tensor y,
+                  torch::Tensor mean, torch::Tensor invstd,
+                  torch::Tensor weight, torch::Tensor bias, int64_t C,
+                  bool relu) {
+  _check_bn_act(x, "x"); _check_bn_act(y, "y");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(weight, "weight"); _check_f32(bias, "bias");
+  const void* res_p = nullptr;
+  if (res.defined() && res.numel()) {
+    _check_bn_act(res, "res");
+    TORCH_CHECK(res.numel() == x.numel(), "residual shape mismatch");
+    res_p = res.data_ptr();
+  }
+  launch_bn_fwd_apply(x.data_ptr(), res_p, y.data_ptr(),
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                      weight.data_ptr<float>(), bias.data_ptr<float>(),
+                      _bn_rows(x, C), (int)C, relu ? 1 : 0,
+                      at::cuda::getCurrentHIPStream().stream());
+}
+
+void bn_bwd_reduce(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
+                   torch::Tensor mean, torch::Tensor invstd,
+                   torch::Tensor accum2, int64_t C, bool relu) {
+  _check_bn_act(dz, "dz"); _check_bn_act(y, "y"); _check_bn_act(x, "x");
+  _check_f32(accum2, "accum2");
+  launch_bn_bwd_reduce(dz.data_ptr(), y.data_ptr(), x.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       accum2.data_ptr<float>(), _bn_rows(x, C), (int)C,
+                       relu ? 1 : 0,
+                       at::cuda::getCurrentHIPStream().stream());
+}
+
+void bn_bwd_finalize(torch::Tensor accum2, torch::Tensor invstd,
+                     torch::Tensor weight, torch::Tensor dweight,
+                     torch::Tensor dbias, torch::Tensor k, int64_t rows,
+                     int64_t C) {
+  _check_f32(accum2, "accum2"); _check_f32(invstd, "invstd");
+  _check_f32(weight, "weight"); _check_f32(dweight, "dweight");
+  _check_f32(dbias, "dbias"); _check_f32(k, "k");
+  TORCH_CHECK(k.numel() == 3 * C, "k must be [3C]");
+  launch_bn_bwd_finalize(accum2.data_ptr<float>(), invstd.data_ptr<float>(),
+                         weight.data_ptr<float>(), dweight.data_ptr<float>(),
+                         dbias.data_ptr<float>(), k.data_ptr<float>(), rows,
+                         (int)C, at::cuda::getCurrentHIPStream().stream());
+}
+
+void bn_bwd_apply(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
+                  torch::Tensor mean, torch::Tensor invstd, torch::Tensor k,
+                  torch::Tensor dx, torch::Tensor dres, int64_t C,
+                  bool relu) {
+  _check_bn_act(dz, "dz"); _check_bn_act(y, "y"); _check_bn_act(x, "x");
+  _check_bn_act(dx, "dx");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(k, "k");
+  void* dres_p = nullptr;
+  if (dres.defined() && dres.numel()) {
+    _check_bn_act(dres, "dres");
+    dres_p = dres.data_ptr();
+  }
+  launch_bn_bwd_apply(dz.data_ptr(), y.data_ptr(), x.data_ptr(),
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                      k.data_ptr<float>(), dx.data_ptr(), dres_p,
+                      _bn_rows(x, C), (int)C, relu ? 1 : 0,
+                      at::cuda::getCurrentHIPStream().stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
@@ -151,4 +289,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "difference + per-group centering of the time-major panel");
   m.def("exog_project_mfma", &exog_project_mfma,
         "design-matrix GEMM beta = P @ Wc on f32 MFMA");
+  m.def("bn_fwd_reduce", &bn_fwd_reduce);
+  m.def("bn_fwd_finalize", &bn_fwd_finalize);
+  m.def("bn_fwd_apply", &bn_fwd_apply);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_finalize", &bn_bwd_finalize);
+  m.def("bn_bwd_apply", &bn_bwd_apply);
 }

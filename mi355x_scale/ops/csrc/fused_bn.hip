@@ -1,0 +1,393 @@
+// Fused BatchNorm(+residual)(+ReLU) for gfx950 (MI355X), NHWC bf16.
+//
+// Replaces the MIOpen spatial-BN path PyTorch-ROCm uses for the ResNet
+// family. Profiled on MI355X (profiles/r01_bench_resnet18_1gpu_kernstats.md),
+// that path is 4 separate kernels per BN (MeanVariance + Norm forward,
+// DScaleDBias + DX backward) run in *fp32* (autocast keeps BN off the
+// bf16 list, inserting bfloat16->float32 copy kernels around every BN),
+// with the adjacent ReLU and residual-add as further standalone
+// elementwise passes — together ~32% of step kernel time.
+//
+// This file fuses the whole thing at bf16 activation width:
+//   fwd:  reduce(x) -> finalize(mean/invstd + running stats) ->
+//         apply: y = [relu](bn(x) [+ residual])            (bf16 in/out)
+//   bwd:  reduce(dz,y,x) -> finalize -> apply: dx, dres    (bf16 in/out)
+// with fp32 accumulation throughout. The ReLU mask is recovered from the
+// saved output y (y > 0), so ReLU backward costs nothing; the residual
+// branch's gradient is the masked upstream gradient and is written by the
+// same backward pass.
+//
+// Layout/mapping (CDNA4-first):
+//   * activations are NHWC ("channels_last"), rows = N*H*W, C channels
+//     contiguous; every load/store is an 8-wide bf16 vector (16 B,
+//     dwordx4) so wavefronts of 64 lanes touch 1 KB per instruction.
+//   * a block of 256 threads (4 wavefronts) covers 256/(C/8) rows per
+//     iteration; grid-stride over rows; grids are sized >> 256 workgroups
+//     where the row count allows, to fill all 8 XCDs.
+//   * block-level partials are tree-reduced through LDS (16 KB of the
+//     160 KB/CU) and committed with fp32 global atomics (C <= 512 means
+//     contention is negligible).
+// All six kernels are stream-ordered device work with no host sync, so
+// the whole fused op captures into hipGraphs (train/graphstep.py).
+//
+// Reference parity: this implements the BatchNorm2d semantics the
+// reference's torchvision resnet50 relies on
+// (deep_learning/2.distributed-data-loading-petastorm.py:150) including
+// running-stat momentum updates and unbiased running variance.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __hip_bfloat16 bf16;
+
+#define VEC 8  // bf16 elements per vector load/store (16 bytes)
+
+union BVec {
+  uint4 u;
+  bf16 h[VEC];
+};
+
+static __device__ __forceinline__ BVec load8(const bf16* p) {
+  BVec v;
+  v.u = *reinterpret_cast<const uint4*>(p);
+  return v;
+}
+
+static __device__ __forceinline__ void store8(bf16* p, const BVec& v) {
+  *reinterpret_cast<uint4*>(p) = v.u;
+}
+
+// ---------------------------------------------------------------- fwd reduce
+// accum: [2*C] fp32, pre-zeroed. accum[c] += sum x_c ; accum[C+c] += sum x_c^2
+__global__ __launch_bounds__(256) void bn_fwd_reduce_kernel(
+    const bf16* __restrict__ x, float* __restrict__ accum,
+    long long rows, int C) {
+  const int lanes = C / VEC;            // vector-lanes per row
+  const int lane = threadIdx.x % lanes; // which 8-channel slot
+  const int rsub = threadIdx.x / lanes; // row within the block's tile
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float s[VEC] = {0.f}, q[VEC] = {0.f};
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    BVec v = load8(x + row * C + (long long)lane * VEC);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float f = __bfloat162float(v.h[j]);
+      s[j] += f;
+      q[j] += f * f;
+    }
+  }
+
+  // Tree-reduce across the rows_per_iter threads sharing each lane.
+  __shared__ float lds[256 * 2 * VEC];
+  float* mys = &lds[(rsub * lanes + lane) * 2 * VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    mys[j] = s[j];
+    mys[VEC + j] = q[j];
+  }
+  __syncthreads();
+  for (int step = rows_per_iter >> 1; step > 0; step >>= 1) {
+    if (rsub < step) {
+      const float* other = &lds[((rsub + step) * lanes + lane) * 2 * VEC];
+#pragma unroll
+      for (int j = 0; j < 2 * VEC; ++j) mys[j] += other[j];
+    }
+    __syncthreads();
+  }
+  if (rsub == 0) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      atomicAdd(&accum[lane * VEC + j], mys[j]);
+      atomicAdd(&accum[C + lane * VEC + j], mys[VEC + j]);
+    }
+  }
+}
+
+// -------------------------------------------------------------- fwd finalize
+// One tiny block: mean/invstd out; momentum update of running stats
+// (unbiased running var, matching torch BatchNorm2d). Device-side so it
+// replays inside hipGraphs.
+__global__ __launch_bounds__(512) void bn_fwd_finalize_kernel(
+    const float* __restrict__ accum, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float momentum, float eps,
+    long long rows, int C, int update_running) {
+  int c = threadIdx.x;
+  if (c >= C) return;
+  float n = (float)rows;
+  float m = accum[c] / n;
+  float var = fmaxf(accum[C + c] / n - m * m, 0.f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (update_running) {
+    float unbiased = rows > 1 ? var * (n / (n - 1.f)) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+// ---------------------------------------------------------------- fwd apply
+// y = relu?( (x - mean)*invstd*w + b  (+ residual) ), all bf16 I/O.
+// Per-thread channel constants are folded once: a = w*invstd,
+// b' = b - mean*a, so the inner loop is one fma per element.
+template <bool RELU, bool RES>
+__global__ __launch_bounds__(256) void bn_fwd_apply_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ res,
+    bf16* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ weight,
+    const float* __restrict__ bias, long long rows, int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float a[VEC], b[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    a[j] = weight[c] * invstd[c];
+    b[j] = bias[c] - mean[c] * a[j];
+  }
+
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec v = load8(x + off);
+    BVec r;
+    if (RES) r = load8(res + off);
+    BVec o;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float f = fmaf(__bfloat162float(v.h[j]), a[j], b[j]);
+      if (RES) f += __bfloat162float(r.h[j]);
+      if (RELU) f = fmaxf(f, 0.f);
+      o.h[j] = __float2bfloat16(f);
+    }
+    store8(y + off, o);
+  }
+}
+
+// ---------------------------------------------------------------- bwd reduce
+// dy = relu-masked upstream grad (mask = saved y > 0).
+// accum2[c] += sum dy ; accum2[C+c] += sum dy * xhat.
+template <bool RELU>
+__global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
+    const bf16* __restrict__ dz, const bf16* __restrict__ y,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ accum2,
+    long long rows, int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float m[VEC], is[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    m[j] = mean[c];
+    is[j] = invstd[c];
+  }
+
+  float sdy[VEC] = {0.f}, sdyx[VEC] = {0.f};
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec g = load8(dz + off);
+    BVec xv = load8(x + off);
+    BVec yv;
+    if (RELU) yv = load8(y + off);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float dy = __bfloat162float(g.h[j]);
+      if (RELU && __bfloat162float(yv.h[j]) <= 0.f) dy = 0.f;
+      float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
+      sdy[j] += dy;
+      sdyx[j] += dy * xhat;
+    }
+  }
+
+  __shared__ float lds[256 * 2 * VEC];
+  float* mys = &lds[(rsub * lanes + lane) * 2 * VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    mys[j] = sdy[j];
+    mys[VEC + j] = sdyx[j];
+  }
+  __syncthreads();
+  for (int step = rows_per_iter >> 1; step > 0; step >>= 1) {
+    if (rsub < step) {
+      const float* other = &lds[((rsub + step) * lanes + lane) * 2 * VEC];
+#pragma unroll
+      for (int j = 0; j < 2 * VEC; ++j) mys[j] += other[j];
+    }
+    __syncthreads();
+  }
+  if (rsub == 0) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      atomicAdd(&accum2[lane * VEC + j], mys[j]);
+      atomicAdd(&accum2[C + lane * VEC + j], mys[VEC + j]);
+    }
+  }
+}
+
+// -------------------------------------------------------------- bwd finalize
+// dweight = sum dy*xhat ; dbias = sum dy ; k = [w*invstd, mean_dy,
+// mean_dy_xhat] per channel for the apply pass.
+__global__ __launch_bounds__(512) void bn_bwd_finalize_kernel(
+    const float* __restrict__ accum2, const float* __restrict__ invstd,
+    const float* __restrict__ weight, float* __restrict__ dweight,
+    float* __restrict__ dbias, float* __restrict__ k, long long rows,
+    int C) {
+  int c = threadIdx.x;
+  if (c >= C) return;
+  float sdy = accum2[c], sdyx = accum2[C + c];
+  dbias[c] = sdy;
+  dweight[c] = sdyx;
+  k[c] = weight[c] * invstd[c];  // the dx scale factor
+  k[C + c] = sdy / (float)rows;
+  k[2 * C + c] = sdyx / (float)rows;
+}
+
+// ---------------------------------------------------------------- bwd apply
+// dx = w*invstd * (dy - mean_dy - xhat*mean_dy_xhat), with xhat
+// recomputed from (x-mean)*invstd and dy relu-masked; dres = dy.
+template <bool RELU, bool RES>
+__global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
+    const bf16* __restrict__ dz, const bf16* __restrict__ y,
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ k,
+    bf16* __restrict__ dx, bf16* __restrict__ dres, long long rows, int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float m[VEC], is[VEC], wv[VEC], k1[VEC], k2[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    m[j] = mean[c];
+    is[j] = invstd[c];
+    wv[j] = k[c];
+    k1[j] = k[C + c];
+    k2[j] = k[2 * C + c];
+  }
+
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec g = load8(dz + off);
+    BVec xv = load8(x + off);
+    BVec yv;
+    if (RELU) yv = load8(y + off);
+    BVec odx, odr;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float dy = __bfloat162float(g.h[j]);
+      if (RELU && __bfloat162float(yv.h[j]) <= 0.f) dy = 0.f;
+      float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
+      float v = wv[j] * (dy - k1[j] - xhat * k2[j]);
+      odx.h[j] = __float2bfloat16(v);
+      if (RES) odr.h[j] = __float2bfloat16(dy);
+    }
+    store8(dx + off, odx);
+    if (RES) store8(dres + off, odr);
+  }
+}
+
+// ------------------------------------------------------------------ launchers
+
+static int pick_grid(long long rows, int C) {
+  // one block advances 256/(C/8) rows per iteration; target >= 2 iters
+  // per block but cap so small late-layer maps don't launch empty blocks.
+  int rows_per_iter = 256 / (C / VEC);
+  long long blocks = (rows + rows_per_iter - 1) / rows_per_iter;
+  if (blocks > 2080) blocks = 2080;  // 8.1 per CU; multiple of 8 XCDs + 1
+  return (int)(blocks > 0 ? blocks : 1);
+}
+
+extern "C" void launch_bn_fwd_reduce(const void* x, float* accum,
+                                     long long rows, int C,
+                                     hipStream_t stream) {
+  hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(pick_grid(rows, C)),
+                     dim3(256), 0, stream, (const bf16*)x, accum, rows, C);
+}
+
+extern "C" void launch_bn_fwd_finalize(const float* accum, float* mean,
+                                       float* invstd, float* running_mean,
+                                       float* running_var, float momentum,
+                                       float eps, long long rows, int C,
+                                       int update_running,
+                                       hipStream_t stream) {
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3(1), dim3(C > 64 ? C : 64),
+                     0, stream, accum, mean, invstd, running_mean,
+                     running_var, momentum, eps, rows, C, update_running);
+}
+
+extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
+                                    const float* mean, const float* invstd,
+                                    const float* weight, const float* bias,
+                                    long long rows, int C, int relu,
+                                    hipStream_t stream) {
+  dim3 grid(pick_grid(rows, C)), block(256);
+#define APPLY(R, S)                                                       \
+  hipLaunchKernelGGL((bn_fwd_apply_kernel<R, S>), grid, block, 0, stream, \
+                     (const bf16*)x, (const bf16*)res, (bf16*)y, mean,    \
+                     invstd, weight, bias, rows, C)
+  if (relu && res) APPLY(true, true);
+  else if (relu) APPLY(true, false);
+  else if (res) APPLY(false, true);
+  else APPLY(false, false);
+#undef APPLY
+}
+
+extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
+                                     const void* x, const float* mean,
+                                     const float* invstd, float* accum2,
+                                     long long rows, int C, int relu,
+                                     hipStream_t stream) {
+  dim3 grid(pick_grid(rows, C)), block(256);
+  if (relu)
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
+                       (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
+                       invstd, accum2, rows, C);
+  else
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
+                       (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
+                       invstd, accum2, rows, C);
+}
+
+extern "C" void launch_bn_bwd_finalize(const float* accum2,
+                                       const float* invstd,
+                                       const float* weight, float* dweight,
+                                       float* dbias, float* k, long long rows,
+                                       int C, hipStream_t stream) {
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(1), dim3(C > 64 ? C : 64),
+                     0, stream, accum2, invstd, weight, dweight, dbias, k,
+                     rows, C);
+}
+
+extern "C" void launch_bn_bwd_apply(const void* dz, const void* y,
+                                    const void* x, const float* mean,
+                                    const float* invstd, const float* k,
+                                    void* dx, void* dres, long long rows,
+                                    int C, int relu, hipStream_t stream) {
+  dim3 grid(pick_grid(rows, C)), block(256);
+#define APPLY(R, S)                                                        \
+  hipLaunchKernelGGL((bn_bwd_apply_kernel<R, S>), grid, block, 0, stream,  \
+                     (const bf16*)dz, (const bf16*)y, (const bf16*)x,      \
+                     mean, invstd, k, (bf16*)dx, (bf16*)dres, rows, C)
+  if (relu && dres) APPLY(true, true);
+  else if (relu) APPLY(true, false);
+  else if (dres) APPLY(false, true);
+  else APPLY(false, false);
+#undef APPLY
+}
