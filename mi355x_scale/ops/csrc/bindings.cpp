@@ -143,11 +143,13 @@ void exog_project_mfma(torch::Tensor P, torch::Tensor wc,
 // ---------------------------------------------------------------- fused BN
 // NHWC bf16 fused BatchNorm(+residual)(+ReLU); see fused_bn.hip.
 
-extern "C" void launch_bn_fwd_reduce(const void* x, float* accum,
-                                     long long rows, int C,
+extern "C" int bn_reduce_blocks(long long rows, int C);
+extern "C" void launch_bn_fwd_reduce(const void* x, float* partial,
+                                     int nblocks, long long rows, int C,
                                      hipStream_t stream);
-extern "C" void launch_bn_fwd_finalize(const float* accum, float* mean,
-                                       float* invstd, float* running_mean,
+extern "C" void launch_bn_fwd_finalize(const float* partial, int nblocks,
+                                       float* mean, float* invstd,
+                                       float* running_mean,
                                        float* running_var, float momentum,
                                        float eps, long long rows, int C,
                                        int update_running, hipStream_t stream);
@@ -158,10 +160,10 @@ extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
                                     hipStream_t stream);
 extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
                                      const void* x, const float* mean,
-                                     const float* invstd, float* accum2,
-                                     long long rows, int C, int relu,
-                                     hipStream_t stream);
-extern "C" void launch_bn_bwd_finalize(const float* accum2,
+                                     const float* invstd, float* partial,
+                                     int nblocks, long long rows, int C,
+                                     int relu, hipStream_t stream);
+extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* invstd,
                                        const float* weight, float* dweight,
                                        float* dbias, float* k, long long rows,
@@ -186,26 +188,30 @@ static long long _bn_rows(const torch::Tensor& x, int64_t C) {
   return (long long)(x.numel() / C);
 }
 
-void bn_fwd_reduce(torch::Tensor x, torch::Tensor accum, int64_t C) {
+torch::Tensor bn_fwd_reduce(torch::Tensor x, int64_t C) {
   _check_bn_act(x, "x");
-  _check_f32(accum, "accum");
-  TORCH_CHECK(accum.numel() == 2 * C, "accum must be [2C]");
-  launch_bn_fwd_reduce(x.data_ptr(), accum.data_ptr<float>(),
-                       _bn_rows(x, C), (int)C,
-                       at::cuda::getCurrentHIPStream().stream());
+  long long rows = _bn_rows(x, C);
+  int nb = bn_reduce_blocks(rows, (int)C);
+  auto partial = torch::empty(
+      {(int64_t)nb * 2 * C},
+      torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+  launch_bn_fwd_reduce(x.data_ptr(), partial.data_ptr<float>(), nb, rows,
+                       (int)C, at::cuda::getCurrentHIPStream().stream());
+  return partial;
 }
 
-void bn_fwd_finalize(torch::Tensor accum, torch::Tensor mean,
+void bn_fwd_finalize(torch::Tensor partial, torch::Tensor mean,
                      torch::Tensor invstd, torch::Tensor running_mean,
                      torch::Tensor running_var, double momentum, double eps,
                      int64_t rows, int64_t C, bool update_running) {
-  _check_f32(accum, "accum"); _check_f32(mean, "mean");
+  _check_f32(partial, "partial"); _check_f32(mean, "mean");
   _check_f32(invstd, "invstd");
   _check_f32(running_mean, "running_mean");
   _check_f32(running_var, "running_var");
-  TORCH_CHECK(C <= 512, "finalize kernel is single-block (C <= 512)");
-  launch_bn_fwd_finalize(accum.data_ptr<float>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(),
+  TORCH_CHECK(partial.numel() % (2 * C) == 0, "partial must be [nb*2C]");
+  int nb = (int)(partial.numel() / (2 * C));
+  launch_bn_fwd_finalize(partial.data_ptr<float>(), nb,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          running_mean.data_ptr<float>(),
                          running_var.data_ptr<float>(), (float)momentum,
                          (float)eps, rows, (int)C, update_running ? 1 : 0,
@@ -232,27 +238,35 @@ void bn_fwd_apply(torch::Tensor x, torch::Tensor res, torch::Tensor y,
                       at::cuda::getCurrentHIPStream().stream());
 }
 
-void bn_bwd_reduce(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
-                   torch::Tensor mean, torch::Tensor invstd,
-                   torch::Tensor accum2, int64_t C, bool relu) {
+torch::Tensor bn_bwd_reduce(torch::Tensor dz, torch::Tensor y,
+                            torch::Tensor x, torch::Tensor mean,
+                            torch::Tensor invstd, int64_t C, bool relu) {
   _check_bn_act(dz, "dz"); _check_bn_act(y, "y"); _check_bn_act(x, "x");
-  _check_f32(accum2, "accum2");
+  long long rows = _bn_rows(x, C);
+  int nb = bn_reduce_blocks(rows, (int)C);
+  auto partial = torch::empty(
+      {(int64_t)nb * 2 * C},
+      torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
   launch_bn_bwd_reduce(dz.data_ptr(), y.data_ptr(), x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       accum2.data_ptr<float>(), _bn_rows(x, C), (int)C,
+                       partial.data_ptr<float>(), nb, rows, (int)C,
                        relu ? 1 : 0,
                        at::cuda::getCurrentHIPStream().stream());
+  return partial;
 }
 
-void bn_bwd_finalize(torch::Tensor accum2, torch::Tensor invstd,
+void bn_bwd_finalize(torch::Tensor partial, torch::Tensor invstd,
                      torch::Tensor weight, torch::Tensor dweight,
                      torch::Tensor dbias, torch::Tensor k, int64_t rows,
                      int64_t C) {
-  _check_f32(accum2, "accum2"); _check_f32(invstd, "invstd");
+  _check_f32(partial, "partial"); _check_f32(invstd, "invstd");
   _check_f32(weight, "weight"); _check_f32(dweight, "dweight");
   _check_f32(dbias, "dbias"); _check_f32(k, "k");
   TORCH_CHECK(k.numel() == 3 * C, "k must be [3C]");
-  launch_bn_bwd_finalize(accum2.data_ptr<float>(), invstd.data_ptr<float>(),
+  TORCH_CHECK(partial.numel() % (2 * C) == 0, "partial must be [nb*2C]");
+  int nb = (int)(partial.numel() / (2 * C));
+  launch_bn_bwd_finalize(partial.data_ptr<float>(), nb,
+                         invstd.data_ptr<float>(),
                          weight.data_ptr<float>(), dweight.data_ptr<float>(),
                          dbias.data_ptr<float>(), k.data_ptr<float>(), rows,
                          (int)C, at::cuda::getCurrentHIPStream().stream());

@@ -58,9 +58,14 @@ static __device__ __forceinline__ void store8(bf16* p, const BVec& v) {
 }
 
 // ---------------------------------------------------------------- fwd reduce
-// accum: [2*C] fp32, pre-zeroed. accum[c] += sum x_c ; accum[C+c] += sum x_c^2
+// Two-stage reduction: every block writes its own [2C] partial slot
+// (partial[block*2C + c] = block-sum x_c, [.. + C + c] = block-sum x_c^2);
+// the finalize kernel sums across blocks. No global atomics — a single
+// fp32 atomic target serializes one RMW per block (measured 406 us/call
+// at 2080 blocks on MI355X, ~20x the memory-bound cost of the pass) and
+// is non-deterministic across graph replays.
 __global__ __launch_bounds__(256) void bn_fwd_reduce_kernel(
-    const bf16* __restrict__ x, float* __restrict__ accum,
+    const bf16* __restrict__ x, float* __restrict__ partial,
     long long rows, int C) {
   const int lanes = C / VEC;            // vector-lanes per row
   const int lane = threadIdx.x % lanes; // which 8-channel slot
@@ -98,28 +103,35 @@ __global__ __launch_bounds__(256) void bn_fwd_reduce_kernel(
     __syncthreads();
   }
   if (rsub == 0) {
+    float* slot = &partial[(long long)blockIdx.x * 2 * C];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      atomicAdd(&accum[lane * VEC + j], mys[j]);
-      atomicAdd(&accum[C + lane * VEC + j], mys[VEC + j]);
+      slot[lane * VEC + j] = mys[j];
+      slot[C + lane * VEC + j] = mys[VEC + j];
     }
   }
 }
 
 // -------------------------------------------------------------- fwd finalize
-// One tiny block: mean/invstd out; momentum update of running stats
+// Sum the per-block partials (thread = channel, coalesced across the
+// block), then mean/invstd out + momentum update of running stats
 // (unbiased running var, matching torch BatchNorm2d). Device-side so it
-// replays inside hipGraphs.
-__global__ __launch_bounds__(512) void bn_fwd_finalize_kernel(
-    const float* __restrict__ accum, float* __restrict__ mean,
-    float* __restrict__ invstd, float* __restrict__ running_mean,
-    float* __restrict__ running_var, float momentum, float eps,
-    long long rows, int C, int update_running) {
-  int c = threadIdx.x;
+// replays inside hipGraphs; fixed summation order = deterministic.
+__global__ __launch_bounds__(256) void bn_fwd_finalize_kernel(
+    const float* __restrict__ partial, int nblocks,
+    float* __restrict__ mean, float* __restrict__ invstd,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float momentum, float eps, long long rows, int C, int update_running) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    s += partial[(long long)b * 2 * C + c];
+    q += partial[(long long)b * 2 * C + C + c];
+  }
   float n = (float)rows;
-  float m = accum[c] / n;
-  float var = fmaxf(accum[C + c] / n - m * m, 0.f);
+  float m = s / n;
+  float var = fmaxf(q / n - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (update_running) {
@@ -172,13 +184,14 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_kernel(
 }
 
 // ---------------------------------------------------------------- bwd reduce
-// dy = relu-masked upstream grad (mask = saved y > 0).
-// accum2[c] += sum dy ; accum2[C+c] += sum dy * xhat.
+// dy = relu-masked upstream grad (mask = saved y > 0). Per-block partial
+// slots (same two-stage scheme as the forward): partial[block*2C + c] =
+// block-sum dy, [.. + C + c] = block-sum dy*xhat.
 template <bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     const bf16* __restrict__ dz, const bf16* __restrict__ y,
     const bf16* __restrict__ x, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ accum2,
+    const float* __restrict__ invstd, float* __restrict__ partial,
     long long rows, int C) {
   const int lanes = C / VEC;
   const int lane = threadIdx.x % lanes;
@@ -229,25 +242,30 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     __syncthreads();
   }
   if (rsub == 0) {
+    float* slot = &partial[(long long)blockIdx.x * 2 * C];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      atomicAdd(&accum2[lane * VEC + j], mys[j]);
-      atomicAdd(&accum2[C + lane * VEC + j], mys[VEC + j]);
+      slot[lane * VEC + j] = mys[j];
+      slot[C + lane * VEC + j] = mys[VEC + j];
     }
   }
 }
 
 // -------------------------------------------------------------- bwd finalize
-// dweight = sum dy*xhat ; dbias = sum dy ; k = [w*invstd, mean_dy,
-// mean_dy_xhat] per channel for the apply pass.
-__global__ __launch_bounds__(512) void bn_bwd_finalize_kernel(
-    const float* __restrict__ accum2, const float* __restrict__ invstd,
-    const float* __restrict__ weight, float* __restrict__ dweight,
-    float* __restrict__ dbias, float* __restrict__ k, long long rows,
-    int C) {
-  int c = threadIdx.x;
+// Cross-block sum of the partials, then: dweight = sum dy*xhat ; dbias =
+// sum dy ; k = [w*invstd, mean_dy, mean_dy_xhat] per channel for apply.
+__global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
+    const float* __restrict__ partial, int nblocks,
+    const float* __restrict__ invstd, const float* __restrict__ weight,
+    float* __restrict__ dweight, float* __restrict__ dbias,
+    float* __restrict__ k, long long rows, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sdy = accum2[c], sdyx = accum2[C + c];
+  float sdy = 0.f, sdyx = 0.f;
+  for (int b = 0; b < nblocks; ++b) {
+    sdy += partial[(long long)b * 2 * C + c];
+    sdyx += partial[(long long)b * 2 * C + C + c];
+  }
   dbias[c] = sdy;
   dweight[c] = sdyx;
   k[c] = weight[c] * invstd[c];  // the dx scale factor
@@ -306,30 +324,41 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
 // ------------------------------------------------------------------ launchers
 
 static int pick_grid(long long rows, int C) {
-  // one block advances 256/(C/8) rows per iteration; target >= 2 iters
-  // per block but cap so small late-layer maps don't launch empty blocks.
+  // one block advances 256/(C/8) rows per iteration; cap so small
+  // late-layer maps don't launch empty blocks.
   int rows_per_iter = 256 / (C / VEC);
   long long blocks = (rows + rows_per_iter - 1) / rows_per_iter;
   if (blocks > 2080) blocks = 2080;  // 8.1 per CU; multiple of 8 XCDs + 1
   return (int)(blocks > 0 ? blocks : 1);
 }
 
-extern "C" void launch_bn_fwd_reduce(const void* x, float* accum,
-                                     long long rows, int C,
-                                     hipStream_t stream) {
-  hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(pick_grid(rows, C)),
-                     dim3(256), 0, stream, (const bf16*)x, accum, rows, C);
+// Reduce grids are capped lower: 512 blocks (2/CU) saturate HBM on a
+// pure-read pass and bound the partial buffer at 512*2C floats.
+extern "C" int bn_reduce_blocks(long long rows, int C) {
+  int rows_per_iter = 256 / (C / VEC);
+  long long blocks = (rows + rows_per_iter - 1) / rows_per_iter;
+  if (blocks > 512) blocks = 512;
+  return (int)(blocks > 0 ? blocks : 1);
 }
 
-extern "C" void launch_bn_fwd_finalize(const float* accum, float* mean,
-                                       float* invstd, float* running_mean,
+extern "C" void launch_bn_fwd_reduce(const void* x, float* partial,
+                                     int nblocks, long long rows, int C,
+                                     hipStream_t stream) {
+  hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(nblocks), dim3(256), 0,
+                     stream, (const bf16*)x, partial, rows, C);
+}
+
+extern "C" void launch_bn_fwd_finalize(const float* partial, int nblocks,
+                                       float* mean, float* invstd,
+                                       float* running_mean,
                                        float* running_var, float momentum,
                                        float eps, long long rows, int C,
                                        int update_running,
                                        hipStream_t stream) {
-  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3(1), dim3(C > 64 ? C : 64),
-                     0, stream, accum, mean, invstd, running_mean,
-                     running_var, momentum, eps, rows, C, update_running);
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, stream, partial, nblocks, mean, invstd,
+                     running_mean, running_var, momentum, eps, rows, C,
+                     update_running);
 }
 
 extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
@@ -351,28 +380,28 @@ extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
 
 extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
                                      const void* x, const float* mean,
-                                     const float* invstd, float* accum2,
-                                     long long rows, int C, int relu,
-                                     hipStream_t stream) {
-  dim3 grid(pick_grid(rows, C)), block(256);
+                                     const float* invstd, float* partial,
+                                     int nblocks, long long rows, int C,
+                                     int relu, hipStream_t stream) {
+  dim3 grid(nblocks), block(256);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
                        (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
-                       invstd, accum2, rows, C);
+                       invstd, partial, rows, C);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
                        (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
-                       invstd, accum2, rows, C);
+                       invstd, partial, rows, C);
 }
 
-extern "C" void launch_bn_bwd_finalize(const float* accum2,
+extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* invstd,
                                        const float* weight, float* dweight,
                                        float* dbias, float* k, long long rows,
                                        int C, hipStream_t stream) {
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(1), dim3(C > 64 ? C : 64),
-                     0, stream, accum2, invstd, weight, dweight, dbias, k,
-                     rows, C);
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256),
+                     dim3(256), 0, stream, partial, nblocks, invstd, weight,
+                     dweight, dbias, k, rows, C);
 }
 
 extern "C" void launch_bn_bwd_apply(const void* dz, const void* y,

@@ -126,6 +126,7 @@ def test_gpu_resnet18_step_runs_fused():
     m = resnet18(num_classes=1000).to(dev).to(
         memory_format=torch.channels_last)
     x = torch.randint(0, 255, (4, 3, 64, 64), device=dev).float()
+    FusedBNReLU2d.gpu_fallbacks.clear()
     with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
         y = m(x)
         loss = y.float().logsumexp(1).mean()
@@ -133,3 +134,6 @@ def test_gpu_resnet18_step_runs_fused():
     assert m.bn1.num_batches_tracked.item() == 1
     assert m.conv1.weight.grad is not None
     assert torch.isfinite(loss)
+    assert not FusedBNReLU2d.gpu_fallbacks, (
+        f"BN layers silently fell back to MIOpen on GPU: "
+        f"{FusedBNReLU2d.gpu_fallbacks[:8]}")
