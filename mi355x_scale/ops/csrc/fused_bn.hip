@@ -103,32 +103,42 @@ __global__ __launch_bounds__(256) void bn_fwd_reduce_kernel(
     __syncthreads();
   }
   if (rsub == 0) {
-    float* slot = &partial[(long long)blockIdx.x * 2 * C];
+    // channel-major partial layout [2C][nblocks]: the finalize wavefront
+    // for channel c then reads a contiguous nblocks-float run.
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      slot[lane * VEC + j] = mys[j];
-      slot[C + lane * VEC + j] = mys[VEC + j];
+      partial[(long long)(lane * VEC + j) * gridDim.x + blockIdx.x] = mys[j];
+      partial[(long long)(C + lane * VEC + j) * gridDim.x + blockIdx.x] =
+          mys[VEC + j];
     }
   }
 }
 
 // -------------------------------------------------------------- fwd finalize
-// Sum the per-block partials (thread = channel, coalesced across the
-// block), then mean/invstd out + momentum update of running stats
-// (unbiased running var, matching torch BatchNorm2d). Device-side so it
-// replays inside hipGraphs; fixed summation order = deterministic.
+// One 64-lane wavefront per channel: lanes stride the [nblocks] partial
+// run (coalesced), then an in-wavefront shuffle tree. mean/invstd out +
+// momentum update of running stats (unbiased running var, matching torch
+// BatchNorm2d). Device-side so it replays inside hipGraphs; fixed
+// summation order = deterministic.
 __global__ __launch_bounds__(256) void bn_fwd_finalize_kernel(
     const float* __restrict__ partial, int nblocks,
     float* __restrict__ mean, float* __restrict__ invstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     float momentum, float eps, long long rows, int C, int update_running) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+  int lane = threadIdx.x % 64;
   if (c >= C) return;
   float s = 0.f, q = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
-    s += partial[(long long)b * 2 * C + c];
-    q += partial[(long long)b * 2 * C + C + c];
+  for (int b = lane; b < nblocks; b += 64) {
+    s += partial[(long long)c * nblocks + b];
+    q += partial[(long long)(C + c) * nblocks + b];
   }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+    q += __shfl_down(q, off, 64);
+  }
+  if (lane != 0) return;
   float n = (float)rows;
   float m = s / n;
   float var = fmaxf(q / n - m * m, 0.f);
@@ -242,30 +252,39 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     __syncthreads();
   }
   if (rsub == 0) {
-    float* slot = &partial[(long long)blockIdx.x * 2 * C];
+    // channel-major partial layout [2C][nblocks] (see forward reduce)
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      slot[lane * VEC + j] = mys[j];
-      slot[C + lane * VEC + j] = mys[VEC + j];
+      partial[(long long)(lane * VEC + j) * gridDim.x + blockIdx.x] = mys[j];
+      partial[(long long)(C + lane * VEC + j) * gridDim.x + blockIdx.x] =
+          mys[VEC + j];
     }
   }
 }
 
 // -------------------------------------------------------------- bwd finalize
-// Cross-block sum of the partials, then: dweight = sum dy*xhat ; dbias =
-// sum dy ; k = [w*invstd, mean_dy, mean_dy_xhat] per channel for apply.
+// One 64-lane wavefront per channel (same scheme as forward finalize):
+// dweight = sum dy*xhat ; dbias = sum dy ; k = [w*invstd, mean_dy,
+// mean_dy_xhat] per channel for the apply pass.
 __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, int nblocks,
     const float* __restrict__ invstd, const float* __restrict__ weight,
     float* __restrict__ dweight, float* __restrict__ dbias,
     float* __restrict__ k, long long rows, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+  int lane = threadIdx.x % 64;
   if (c >= C) return;
   float sdy = 0.f, sdyx = 0.f;
-  for (int b = 0; b < nblocks; ++b) {
-    sdy += partial[(long long)b * 2 * C + c];
-    sdyx += partial[(long long)b * 2 * C + C + c];
+  for (int b = lane; b < nblocks; b += 64) {
+    sdy += partial[(long long)c * nblocks + b];
+    sdyx += partial[(long long)(C + c) * nblocks + b];
   }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sdy += __shfl_down(sdy, off, 64);
+    sdyx += __shfl_down(sdyx, off, 64);
+  }
+  if (lane != 0) return;
   dbias[c] = sdy;
   dweight[c] = sdyx;
   k[c] = weight[c] * invstd[c];  // the dx scale factor
@@ -355,7 +374,8 @@ extern "C" void launch_bn_fwd_finalize(const float* partial, int nblocks,
                                        float eps, long long rows, int C,
                                        int update_running,
                                        hipStream_t stream) {
-  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 255) / 256),
+  // 4 wavefronts (= 4 channels) per block
+  hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((C + 3) / 4),
                      dim3(256), 0, stream, partial, nblocks, mean, invstd,
                      running_mean, running_var, momentum, eps, rows, C,
                      update_running);
@@ -399,7 +419,7 @@ extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* weight, float* dweight,
                                        float* dbias, float* k, long long rows,
                                        int C, hipStream_t stream) {
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256),
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 3) / 4),
                      dim3(256), 0, stream, partial, nblocks, invstd, weight,
                      dweight, dbias, k, rows, C);
 }

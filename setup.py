@@ -10,7 +10,8 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+from torch.utils.cpp_extension import (BuildExtension, CppExtension,
+                                       CUDAExtension)
 
 SRC = [
     "mi355x_scale/ops/csrc/bindings.cpp",
@@ -32,7 +33,14 @@ setup(
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        # host-side C++ group-gather engine (no HIP): multithreaded key
+        # factorize + panel scatter
+        CppExtension(
+            name="mi355x_scale.groupby._gather",
+            sources=["mi355x_scale/groupby/csrc/gather.cpp"],
+            extra_compile_args={"cxx": ["-O3"]},
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
