@@ -125,13 +125,13 @@ def run_fine_grained_forecast_gpu(demand_df: pd.DataFrame,
     """
     from ..groupby.gather import long_from_panel, panel_from_long
     from .batched import batched_fit_gpu
-    if not set(EXO_COLS) <= set(demand_df.columns):
-        demand_df = add_exo_variables(demand_df)
-    # vectorized group gather (the Spark-shuffle replacement): one scatter
+    # C++ group gather (the Spark-shuffle replacement): factorize + scatter
     y, gindex, dates = panel_from_long(demand_df, ["Product", "SKU"],
                                        "Date", "Demand")
-    one = demand_df[demand_df["SKU"] == demand_df["SKU"].iloc[0]]
-    exog = one.sort_values("Date")[EXO_COLS].to_numpy()
+    # exog indicators are a pure function of the week — compute on the T
+    # unique dates, not by enriching/scanning the full long frame
+    exog = add_exo_variables(
+        pd.DataFrame({"Date": pd.to_datetime(dates)}))[EXO_COLS].to_numpy()
     T = y.shape[1]
     out = batched_fit_gpu(y, exog, orders or DEFAULT_GPU_ORDERS,
                           train_len=T - horizon)
