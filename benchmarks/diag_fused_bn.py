@@ -37,7 +37,15 @@ def main():
     print(f"eager exec fallbacks: {len(FusedBNReLU2d.gpu_fallbacks)}")
     for f in FusedBNReLU2d.gpu_fallbacks[:25]:
         print("  ", f)
+    # any leftover nn.BatchNorm2d is a silent MIOpen path
+    import torch.nn as nn
+    leftovers = [n for n, m in model.named_modules()
+                 if isinstance(m, nn.BatchNorm2d)]
+    print(f"plain nn.BatchNorm2d modules: {leftovers}")
 
+    # drop the eager autograd graph before capturing (a live
+    # AccumulateGrad from a prior iteration breaks graph capture)
+    del loss
     FusedBNReLU2d.gpu_fallbacks.clear()
     for p in model.parameters():
         p.grad = None

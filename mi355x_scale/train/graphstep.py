@@ -47,7 +47,13 @@ class GraphedTrainStep:
                                       dtype=torch.float32)
         off = 0
         for p in params:
-            p.grad = self.flat_grads[off:off + p.numel()].view_as(p)
+            # grad views adopt each param's own (dense) stride order —
+            # channels-last conv weights get channels-last grad views, so
+            # backward accumulates without a per-weight layout permute
+            # (the "gradient layout contract") and the flat all-reduce
+            # still sums identical byte layouts across ranks.
+            p.grad = self.flat_grads[off:off + p.numel()].as_strided(
+                p.shape, p.stride())
             off += p.numel()
 
         def _fwd_bwd():

@@ -34,6 +34,17 @@ setup(
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
             },
         ),
+        # hand-written xGMI p2p primitives (IPC buffers + multi-source
+        # reduce) behind parallel/p2p_allreduce.py
+        CUDAExtension(
+            name="mi355x_scale.parallel._p2p",
+            sources=["mi355x_scale/parallel/csrc/p2p_bindings.cpp",
+                     "mi355x_scale/parallel/csrc/p2p.hip"],
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950"],
+            },
+        ),
         # host-side C++ group-gather engine (no HIP): multithreaded key
         # factorize + panel scatter
         CppExtension(

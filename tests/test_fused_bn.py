@@ -42,6 +42,20 @@ def test_cpu_residual_relu_composition():
     torch.testing.assert_close(y, want)
 
 
+def test_no_plain_batchnorm_in_models():
+    """Every BN in the model family must be the fused module — a plain
+    nn.BatchNorm2d silently routes to MIOpen on GPU (this caught the
+    stem/downsample BNs left unconverted in round 1)."""
+    from mi355x_scale.models import resnet18, resnet50
+    for model in (resnet18(num_classes=10), resnet50(num_classes=10)):
+        leftovers = [n for n, m in model.named_modules()
+                     if isinstance(m, torch.nn.BatchNorm2d)]
+        assert not leftovers, leftovers
+        fused = [m for m in model.modules()
+                 if isinstance(m, FusedBNReLU2d)]
+        assert len(fused) >= 20
+
+
 def test_state_dict_interop_with_batchnorm2d():
     m = FusedBNReLU2d(32)
     bn = torch.nn.BatchNorm2d(32)
