@@ -110,8 +110,10 @@ def main() -> None:
     use_graph = use_cuda and not args.no_graph
     runner = model
     if use_graph:
-        optimizer = torch.optim.Adam(model.parameters(), lr=1e-5,
-                                     foreach=True, capturable=True)
+        # fused flat Adam: one HIP kernel per optimizer step, and its
+        # flat grad buffer doubles as the single all-reduce target
+        from mi355x_scale.train.flat_adam import FlatAdam
+        optimizer = FlatAdam(model.parameters(), lr=1e-5)
     else:
         if n_gpus > 1:
             kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,

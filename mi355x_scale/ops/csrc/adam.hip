@@ -1,0 +1,83 @@
+// Fused flat Adam for gfx950: the whole optimizer step in one
+// memory-bound pass.
+//
+// The reference's optimizer is torch.optim.Adam (deep_learning/
+// 2.distributed-data-loading-petastorm.py:158-161). PyTorch's
+// capturable foreach Adam replays ~70 small elementwise kernels per
+// step (div/addcmul/lerp per parameter-group chunk) — ~1.2 ms/step for
+// ResNet-18 on MI355X. With parameters, gradients and both moments laid
+// out as flat fp32 buffers (train/graphstep.py owns that layout
+// already), Adam is one elementwise kernel over 4 streams of data:
+// ~315 MB of HBM traffic, ~45 us at the 8 TB/s roofline.
+//
+// Bias correction uses a device step counter (incremented by a
+// single-lane prologue kernel) so the whole step replays inside
+// hipGraphs with no host-side state.
+
+#include <hip/hip_runtime.h>
+
+__global__ void adam_bump_step_kernel(int* step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
+}
+
+__global__ __launch_bounds__(256) void adam_step_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v,
+    const int* __restrict__ step, float lr, float beta1, float beta2,
+    float eps, float weight_decay, long long n) {
+  const int t = *step;
+  const float bc1 = 1.f - powf(beta1, (float)t);
+  const float bc2 = 1.f - powf(beta2, (float)t);
+  const float step_size = lr / bc1;
+  const long long stride = (long long)gridDim.x * blockDim.x * 4;
+
+  for (long long base = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       base < n; base += stride) {
+    // float4 fast path; scalar tail handled by the same loop body
+    const int lanes = (int)min((long long)4, n - base);
+    if (lanes == 4) {
+      float4 pv = *reinterpret_cast<float4*>(p + base);
+      float4 gv = *reinterpret_cast<const float4*>(g + base);
+      float4 mv = *reinterpret_cast<float4*>(m + base);
+      float4 vv = *reinterpret_cast<float4*>(v + base);
+      float* pp = reinterpret_cast<float*>(&pv);
+      const float* gp = reinterpret_cast<const float*>(&gv);
+      float* mp = reinterpret_cast<float*>(&mv);
+      float* vp = reinterpret_cast<float*>(&vv);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float grad = gp[j] + weight_decay * pp[j];
+        mp[j] = beta1 * mp[j] + (1.f - beta1) * grad;
+        vp[j] = beta2 * vp[j] + (1.f - beta2) * grad * grad;
+        float denom = sqrtf(vp[j] / bc2) + eps;
+        pp[j] -= step_size * mp[j] / denom;
+      }
+      *reinterpret_cast<float4*>(p + base) = pv;
+      *reinterpret_cast<float4*>(m + base) = mv;
+      *reinterpret_cast<float4*>(v + base) = vv;
+    } else {
+      for (int j = 0; j < lanes; ++j) {
+        float grad = g[base + j] + weight_decay * p[base + j];
+        float mj = beta1 * m[base + j] + (1.f - beta1) * grad;
+        float vj = beta2 * v[base + j] + (1.f - beta2) * grad * grad;
+        m[base + j] = mj;
+        v[base + j] = vj;
+        p[base + j] -= step_size * mj / (sqrtf(vj / bc2) + eps);
+      }
+    }
+  }
+}
+
+extern "C" void launch_adam_step(float* p, const float* g, float* m,
+                                 float* v, int* step, float lr, float beta1,
+                                 float beta2, float eps, float weight_decay,
+                                 long long n, hipStream_t stream) {
+  hipLaunchKernelGGL(adam_bump_step_kernel, dim3(1), dim3(64), 0, stream,
+                     step);
+  const int block = 256;
+  long long want = (n + block * 4 - 1) / (block * 4);
+  int grid = (int)(want < 1 ? 1 : (want > 2080 ? 2080 : want));
+  hipLaunchKernelGGL(adam_step_kernel, dim3(grid), dim3(block), 0, stream,
+                     p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,
+                     n);
+}

@@ -292,6 +292,32 @@ void bn_bwd_apply(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
                       at::cuda::getCurrentHIPStream().stream());
 }
 
+// ---------------------------------------------------------------- flat Adam
+extern "C" void launch_adam_step(float* p, const float* g, float* m,
+                                 float* v, int* step, float lr, float beta1,
+                                 float beta2, float eps, float weight_decay,
+                                 long long n, hipStream_t stream);
+
+// One fused Adam step over the flat fp32 param/grad/moment buffers.
+// step: int32 device scalar (bumped on-device; hipGraph-replayable).
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor step, double lr, double beta1,
+               double beta2, double eps, double weight_decay) {
+  _check_f32(p, "p"); _check_f32(g, "g"); _check_f32(m, "m");
+  _check_f32(v, "v");
+  TORCH_CHECK(step.is_cuda() && step.scalar_type() == torch::kInt32 &&
+                  step.numel() == 1,
+              "step must be a 1-element int32 device tensor");
+  long long n = p.numel();
+  TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n,
+              "flat buffer length mismatch");
+  launch_adam_step(p.data_ptr<float>(), g.data_ptr<float>(),
+                   m.data_ptr<float>(), v.data_ptr<float>(),
+                   step.data_ptr<int>(), (float)lr, (float)beta1,
+                   (float)beta2, (float)eps, (float)weight_decay, n,
+                   at::cuda::getCurrentHIPStream().stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
@@ -309,4 +335,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_finalize", &bn_bwd_finalize);
   m.def("bn_bwd_apply", &bn_bwd_apply);
+  m.def("adam_step", &adam_step,
+        "fused flat Adam step (one kernel over p/g/m/v)");
 }

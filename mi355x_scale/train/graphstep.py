@@ -40,21 +40,27 @@ class GraphedTrainStep:
         }
 
         # Flat gradient buffer; every p.grad is a view into it so the
-        # cross-rank all-reduce is ONE call.
-        params = [p for p in model.parameters() if p.requires_grad]
-        total = sum(p.numel() for p in params)
-        self.flat_grads = torch.zeros(total, device=self.device,
-                                      dtype=torch.float32)
-        off = 0
-        for p in params:
-            # grad views adopt each param's own (dense) stride order —
-            # channels-last conv weights get channels-last grad views, so
-            # backward accumulates without a per-weight layout permute
-            # (the "gradient layout contract") and the flat all-reduce
-            # still sums identical byte layouts across ranks.
-            p.grad = self.flat_grads[off:off + p.numel()].as_strided(
-                p.shape, p.stride())
-            off += p.numel()
+        # cross-rank all-reduce is ONE call. FlatAdam (train/flat_adam.py)
+        # already owns such a buffer and has the views installed — reuse
+        # it so optimizer + all-reduce share one layout.
+        if hasattr(optimizer, "flat_grads"):
+            self.flat_grads = optimizer.flat_grads
+        else:
+            params = [p for p in model.parameters() if p.requires_grad]
+            total = sum(p.numel() for p in params)
+            self.flat_grads = torch.zeros(total, device=self.device,
+                                          dtype=torch.float32)
+            off = 0
+            for p in params:
+                # grad views adopt each param's own (dense) stride order —
+                # channels-last conv weights get channels-last grad views,
+                # so backward accumulates without a per-weight layout
+                # permute (the "gradient layout contract") and the flat
+                # all-reduce still sums identical byte layouts across
+                # ranks.
+                p.grad = self.flat_grads[off:off + p.numel()].as_strided(
+                    p.shape, p.stride())
+                off += p.numel()
 
         def _fwd_bwd():
             self.flat_grads.zero_()

@@ -19,6 +19,7 @@ SRC = [
     "mi355x_scale/ops/csrc/groupfit.hip",
     "mi355x_scale/ops/csrc/mfma_project.hip",
     "mi355x_scale/ops/csrc/fused_bn.hip",
+    "mi355x_scale/ops/csrc/adam.hip",
 ]
 
 setup(
