@@ -40,8 +40,18 @@ class FlatAdam:
 
         self.flat_params = torch.empty(total, dtype=torch.float32,
                                        device=device)
-        self.flat_grads = torch.zeros(total, dtype=torch.float32,
-                                      device=device)
+        import os
+        if (device.type == "cuda"
+                and os.environ.get("MI355X_P2P_ALLREDUCE") == "1"):
+            # IPC-shareable grad buffer so the hand-written xGMI p2p
+            # all-reduce (parallel/p2p_allreduce.py) can map it from
+            # peer ranks; hipIpc handles need the hipMalloc base pointer.
+            from ..parallel.p2p_allreduce import alloc_shared
+            self.flat_grads = alloc_shared(total, device)
+            self.flat_grads.zero_()
+        else:
+            self.flat_grads = torch.zeros(total, dtype=torch.float32,
+                                          device=device)
         off = 0
         for p in self.params:
             n = p.numel()

@@ -30,6 +30,7 @@ class GraphedTrainStep:
         self.model = model
         self.optimizer = optimizer
         self.world_size = world_size
+        self._p2p = None  # lazy MI355X_P2P_ALLREDUCE=1 path
         self.device = next(model.parameters()).device
         assert self.device.type == "cuda", "graph capture needs a GPU"
         self.autocast_dtype = autocast_dtype
@@ -117,8 +118,18 @@ class GraphedTrainStep:
             h = self.flat_grads.cpu()
             dist.all_reduce(h)
             self.flat_grads.copy_(h)
-        else:
-            dist.all_reduce(self.flat_grads)  # RCCL over xGMI
+            return
+        import os
+        if os.environ.get("MI355X_P2P_ALLREDUCE") == "1":
+            # hand-written direct xGMI path (7 concurrent link reads vs
+            # the ring's single-link bound); opt-in, needs the flat
+            # buffer from alloc_shared (FlatAdam handles that)
+            if self._p2p is None:
+                from ..parallel.p2p_allreduce import P2PAllReduce
+                self._p2p = P2PAllReduce(self.flat_grads)
+            self._p2p.all_reduce_()
+            return
+        dist.all_reduce(self.flat_grads)  # RCCL over xGMI
 
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
         for k, v in batch.items():

@@ -121,7 +121,14 @@ class Trainer:
             # bucket hooks fire on the training forward pass.
             wrapped = DDP(_TrainStepShim(model), **kwargs)
 
-        optimizer = model.configure_optimizers()
+        if graph_on:
+            # fused flat Adam: one HIP kernel per step; its flat grad
+            # buffer is the graphed step's single all-reduce target
+            from .flat_adam import FlatAdam
+            optimizer = FlatAdam(model.parameters(),
+                                 lr=getattr(model, "lr", 1e-3))
+        else:
+            optimizer = model.configure_optimizers()
         start_epoch = 0
         if self.resume_from:
             state = CheckpointManager.load(self.resume_from, model,
