@@ -29,9 +29,12 @@ from . import _C, HAVE_EXT, require_ext
 
 
 def _hip_supported(x: torch.Tensor) -> bool:
+    # C = 8*2^k up to 2048: one 256-thread block must tile whole rows
+    # (256 % lanes == 0) and a row must fit one block (lanes <= 256).
+    # Covers every BN in the ResNet family (64..2048).
     c = x.shape[1]
     return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4
-            and c % 8 == 0 and c <= 512 and (256 % (c // 8)) == 0)
+            and c % 8 == 0 and c <= 2048 and (256 % (c // 8)) == 0)
 
 
 class _FusedBNFn(torch.autograd.Function):

@@ -64,7 +64,8 @@ def test_state_dict_interop_with_batchnorm2d():
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("C,hw", [(64, 56), (128, 28), (512, 7)])
+@pytest.mark.parametrize("C,hw", [(64, 56), (128, 28), (512, 7),
+                                  (2048, 4)])
 @pytest.mark.parametrize("relu,use_res", [(True, False), (False, False),
                                           (True, True)])
 def test_gpu_fused_fwd_bwd_vs_fp32(C, hw, relu, use_res):
@@ -129,6 +130,26 @@ def test_gpu_eval_mode_uses_running_stats():
                         m.running_mean.clone(), m.running_var.clone(),
                         False, 0.1, m.eps, True)
     torch.testing.assert_close(y.float(), want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.gpu
+def test_gpu_resnet50_bottleneck_step():
+    """Bottleneck blocks (reference's actual model family,
+    deep_learning/2...py:150) through the fused BN path: one fwd+bwd,
+    zero fallbacks, finite loss."""
+    from mi355x_scale.models import resnet50
+    dev = torch.device("cuda:0")
+    m = resnet50(num_classes=100).to(dev).to(
+        memory_format=torch.channels_last)
+    x = torch.randn(4, 3, 64, 64, device=dev)
+    FusedBNReLU2d.gpu_fallbacks.clear()
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        loss = m(x).float().logsumexp(1).mean()
+    loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    assert not FusedBNReLU2d.gpu_fallbacks, \
+        FusedBNReLU2d.gpu_fallbacks[:8]
 
 
 @pytest.mark.gpu
