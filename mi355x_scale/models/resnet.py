@@ -18,6 +18,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.fused_bn import FusedBNReLU2d
+from ..ops.maxpool import MaxPool3x3s2
 
 
 def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -75,7 +76,7 @@ class ResNet(nn.Module):
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = FusedBNReLU2d(64)  # stem BN+ReLU in one pass
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.maxpool = MaxPool3x3s2()  # u8-code NHWC pool kernel on GPU
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)

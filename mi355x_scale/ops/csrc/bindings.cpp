@@ -292,6 +292,44 @@ void bn_bwd_apply(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
                       at::cuda::getCurrentHIPStream().stream());
 }
 
+// ----------------------------------------------------------------- maxpool
+extern "C" void launch_maxpool_fwd(const void* x, void* y,
+                                   unsigned char* code, int N, int H, int W,
+                                   int Ho, int Wo, int C,
+                                   hipStream_t stream);
+extern "C" void launch_maxpool_bwd(const void* dy,
+                                   const unsigned char* code, void* dx,
+                                   int N, int H, int W, int Ho, int Wo,
+                                   int C, hipStream_t stream);
+
+// x: bf16 channels_last [N,C,H,W]; y: [N,C,Ho,Wo]; code: u8 with y's
+// NHWC memory order. 3x3/stride-2/pad-1 only (the ResNet stem pool).
+void maxpool3x3s2_fwd(torch::Tensor x, torch::Tensor y,
+                      torch::Tensor code) {
+  _check_bn_act(x, "x"); _check_bn_act(y, "y");
+  TORCH_CHECK(code.scalar_type() == torch::kUInt8 && code.is_cuda());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Ho = y.size(2), Wo = y.size(3);
+  TORCH_CHECK(C % 8 == 0 && 256 % (C / 8) == 0, "bad C");
+  TORCH_CHECK(Ho == (H + 1) / 2 && Wo == (W + 1) / 2, "3x3 s2 p1 shape");
+  TORCH_CHECK(code.numel() == y.numel());
+  launch_maxpool_fwd(x.data_ptr(), y.data_ptr(), code.data_ptr<uint8_t>(),
+                     N, H, W, Ho, Wo, C,
+                     at::cuda::getCurrentHIPStream().stream());
+}
+
+void maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor code,
+                      torch::Tensor dx) {
+  _check_bn_act(dy, "dy"); _check_bn_act(dx, "dx");
+  TORCH_CHECK(code.scalar_type() == torch::kUInt8 && code.is_cuda());
+  int N = dx.size(0), C = dx.size(1), H = dx.size(2), W = dx.size(3);
+  int Ho = dy.size(2), Wo = dy.size(3);
+  TORCH_CHECK(code.numel() == dy.numel());
+  launch_maxpool_bwd(dy.data_ptr(), code.data_ptr<uint8_t>(),
+                     dx.data_ptr(), N, H, W, Ho, Wo, C,
+                     at::cuda::getCurrentHIPStream().stream());
+}
+
 // ---------------------------------------------------------------- flat Adam
 extern "C" void launch_adam_step(float* p, const float* g, float* m,
                                  float* v, int* step, float lr, float beta1,
@@ -337,4 +375,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("adam_step", &adam_step,
         "fused flat Adam step (one kernel over p/g/m/v)");
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
 }

@@ -20,6 +20,7 @@ SRC = [
     "mi355x_scale/ops/csrc/mfma_project.hip",
     "mi355x_scale/ops/csrc/fused_bn.hip",
     "mi355x_scale/ops/csrc/adam.hip",
+    "mi355x_scale/ops/csrc/maxpool.hip",
 ]
 
 setup(
