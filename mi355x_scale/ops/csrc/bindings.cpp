@@ -161,8 +161,13 @@ extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
 extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
                                      const void* x, const float* mean,
                                      const float* invstd, float* partial,
-                                     int nblocks, long long rows, int C,
-                                     int relu, hipStream_t stream);
+                                     void* dym, int nblocks, long long rows,
+                                     int C, int relu, hipStream_t stream);
+extern "C" void launch_bn_bwd_apply_dym(const void* dym, const void* x,
+                                        const float* mean,
+                                        const float* invstd, const float* k,
+                                        void* dx, long long rows, int C,
+                                        hipStream_t stream);
 extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* invstd,
                                        const float* weight, float* dweight,
@@ -238,10 +243,20 @@ void bn_fwd_apply(torch::Tensor x, torch::Tensor res, torch::Tensor y,
                       at::cuda::getCurrentHIPStream().stream());
 }
 
+// dym (optional, pass an empty tensor to skip): receives the relu-masked
+// upstream gradient — for residual layers it IS the residual grad and
+// feeds bn_bwd_apply_dym.
 torch::Tensor bn_bwd_reduce(torch::Tensor dz, torch::Tensor y,
                             torch::Tensor x, torch::Tensor mean,
-                            torch::Tensor invstd, int64_t C, bool relu) {
+                            torch::Tensor invstd, torch::Tensor dym,
+                            int64_t C, bool relu) {
   _check_bn_act(dz, "dz"); _check_bn_act(y, "y"); _check_bn_act(x, "x");
+  void* dym_p = nullptr;
+  if (dym.defined() && dym.numel()) {
+    _check_bn_act(dym, "dym");
+    TORCH_CHECK(dym.numel() == x.numel(), "dym shape mismatch");
+    dym_p = dym.data_ptr();
+  }
   long long rows = _bn_rows(x, C);
   int nb = bn_reduce_blocks(rows, (int)C);
   auto partial = torch::empty(
@@ -249,10 +264,23 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dz, torch::Tensor y,
       torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
   launch_bn_bwd_reduce(dz.data_ptr(), y.data_ptr(), x.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       partial.data_ptr<float>(), nb, rows, (int)C,
+                       partial.data_ptr<float>(), dym_p, nb, rows, (int)C,
                        relu ? 1 : 0,
                        at::cuda::getCurrentHIPStream().stream());
   return partial;
+}
+
+void bn_bwd_apply_dym(torch::Tensor dym, torch::Tensor x,
+                      torch::Tensor mean, torch::Tensor invstd,
+                      torch::Tensor k, torch::Tensor dx, int64_t C) {
+  _check_bn_act(dym, "dym"); _check_bn_act(x, "x"); _check_bn_act(dx, "dx");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(k, "k");
+  launch_bn_bwd_apply_dym(dym.data_ptr(), x.data_ptr(),
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          k.data_ptr<float>(), dx.data_ptr(),
+                          _bn_rows(x, C), (int)C,
+                          at::cuda::getCurrentHIPStream().stream());
 }
 
 void bn_bwd_finalize(torch::Tensor partial, torch::Tensor invstd,
@@ -373,6 +401,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_finalize", &bn_bwd_finalize);
   m.def("bn_bwd_apply", &bn_bwd_apply);
+  m.def("bn_bwd_apply_dym", &bn_bwd_apply_dym);
   m.def("adam_step", &adam_step,
         "fused flat Adam step (one kernel over p/g/m/v)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);

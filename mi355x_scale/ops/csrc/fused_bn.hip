@@ -195,14 +195,16 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_kernel(
 
 // ---------------------------------------------------------------- bwd reduce
 // dy = relu-masked upstream grad (mask = saved y > 0). Per-block partial
-// slots (same two-stage scheme as the forward): partial[block*2C + c] =
-// block-sum dy, [.. + C + c] = block-sum dy*xhat.
-template <bool RELU>
+// slots (same two-stage scheme as the forward). When WRITE_DY, the
+// masked dy is also materialized to dym: for residual layers dym IS the
+// residual-branch gradient, and the apply pass then reads (dym, x)
+// instead of (dz, y, x) — one read and one write less per layer.
+template <bool RELU, bool WRITE_DY>
 __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     const bf16* __restrict__ dz, const bf16* __restrict__ y,
     const bf16* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ partial,
-    long long rows, int C) {
+    bf16* __restrict__ dym, long long rows, int C) {
   const int lanes = C / VEC;
   const int lane = threadIdx.x % lanes;
   const int rsub = threadIdx.x / lanes;
@@ -225,6 +227,7 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     BVec xv = load8(x + off);
     BVec yv;
     if (RELU) yv = load8(y + off);
+    BVec dyv;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float dy = __bfloat162float(g.h[j]);
@@ -232,7 +235,9 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
       float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
       sdy[j] += dy;
       sdyx[j] += dy * xhat;
+      if (WRITE_DY) dyv.h[j] = __float2bfloat16(dy);
     }
+    if (WRITE_DY) store8(dym + off, dyv);
   }
 
   __shared__ float lds[256 * 2 * VEC];
@@ -290,6 +295,47 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
   k[c] = weight[c] * invstd[c];  // the dx scale factor
   k[C + c] = sdy / (float)rows;
   k[2 * C + c] = sdyx / (float)rows;
+}
+
+// ------------------------------------------------------- bwd apply (dym)
+// Residual-layer variant: reads the pre-masked dym written by the
+// reduce pass (which doubles as the residual gradient) + x; writes dx.
+__global__ __launch_bounds__(256) void bn_bwd_apply_dym_kernel(
+    const bf16* __restrict__ dym, const bf16* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ k, bf16* __restrict__ dx, long long rows,
+    int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float m[VEC], is[VEC], wv[VEC], k1[VEC], k2[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    m[j] = mean[c];
+    is[j] = invstd[c];
+    wv[j] = k[c];
+    k1[j] = k[C + c];
+    k2[j] = k[2 * C + c];
+  }
+
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec g = load8(dym + off);
+    BVec xv = load8(x + off);
+    BVec odx;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float dy = __bfloat162float(g.h[j]);
+      float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
+      odx.h[j] = __float2bfloat16(wv[j] * (dy - k1[j] - xhat * k2[j]));
+    }
+    store8(dx + off, odx);
+  }
 }
 
 // ---------------------------------------------------------------- bwd apply
@@ -401,17 +447,28 @@ extern "C" void launch_bn_fwd_apply(const void* x, const void* res, void* y,
 extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
                                      const void* x, const float* mean,
                                      const float* invstd, float* partial,
-                                     int nblocks, long long rows, int C,
-                                     int relu, hipStream_t stream) {
+                                     void* dym, int nblocks, long long rows,
+                                     int C, int relu, hipStream_t stream) {
   dim3 grid(nblocks), block(256);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
-                       (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
-                       invstd, partial, rows, C);
-  else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
-                       (const bf16*)dz, (const bf16*)y, (const bf16*)x, mean,
-                       invstd, partial, rows, C);
+#define REDUCE(R, W)                                                       \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, W>), grid, block, 0, stream, \
+                     (const bf16*)dz, (const bf16*)y, (const bf16*)x,      \
+                     mean, invstd, partial, (bf16*)dym, rows, C)
+  if (relu && dym) REDUCE(true, true);
+  else if (relu) REDUCE(true, false);
+  else if (dym) REDUCE(false, true);
+  else REDUCE(false, false);
+#undef REDUCE
+}
+
+extern "C" void launch_bn_bwd_apply_dym(const void* dym, const void* x,
+                                        const float* mean,
+                                        const float* invstd, const float* k,
+                                        void* dx, long long rows, int C,
+                                        hipStream_t stream) {
+  hipLaunchKernelGGL(bn_bwd_apply_dym_kernel, dim3(pick_grid(rows, C)),
+                     dim3(256), 0, stream, (const bf16*)dym, (const bf16*)x,
+                     mean, invstd, k, (bf16*)dx, rows, C);
 }
 
 extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,

@@ -68,14 +68,27 @@ class _FusedBNFn(torch.autograd.Function):
         k = torch.empty(3 * C, dtype=torch.float32, device=dev)
         dweight = torch.empty(C, dtype=torch.float32, device=dev)
         dbias = torch.empty(C, dtype=torch.float32, device=dev)
-        partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, C, ctx.relu)
+        dx = torch.empty_like(x)
+        if ctx.has_res:
+            # residual layer: the reduce pass materializes the masked
+            # upstream grad (dym) which IS the residual gradient; the
+            # apply pass then reads (dym, x) — one read+write less than
+            # re-deriving dy from (dz, y).
+            dym = torch.empty_like(x)
+            partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, dym, C,
+                                       ctx.relu)
+            _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
+                               rows, C)
+            _C.bn_bwd_apply_dym(dym, x, mean, invstd, k, dx, C)
+            return (dx, dym, dweight, dbias,
+                    None, None, None, None, None)
+        partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, x.new_empty(0),
+                                   C, ctx.relu)
         _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
                            rows, C)
-        dx = torch.empty_like(x)
-        dres = torch.empty_like(x) if ctx.has_res else x.new_empty(0)
-        _C.bn_bwd_apply(dz, y, x, mean, invstd, k, dx, dres, C, ctx.relu)
-        return (dx, dres if ctx.has_res else None, dweight, dbias,
-                None, None, None, None, None)
+        _C.bn_bwd_apply(dz, y, x, mean, invstd, k, dx, x.new_empty(0), C,
+                        ctx.relu)
+        return (dx, None, dweight, dbias, None, None, None, None, None)
 
 
 class FusedBNReLU2d(nn.Module):
