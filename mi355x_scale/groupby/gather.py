@@ -137,16 +137,34 @@ def panel_from_long(df: pd.DataFrame, keys: Sequence[str], time_col: str,
 
 def long_from_panel(panel: np.ndarray, gindex, tvals,
                     keys: Sequence[str], time_col: str,
-                    value_cols: List[Tuple[str, np.ndarray]]
-                    ) -> pd.DataFrame:
-    """Inverse: [G,T] matrices back to a long frame (keys × time rows)."""
+                    value_cols: List[Tuple[str, np.ndarray]],
+                    categorical_keys: bool = True) -> pd.DataFrame:
+    """Inverse: [G,T] matrices back to a long frame (keys × time rows).
+
+    Key columns are emitted as pandas Categoricals by default: the
+    G·T-row frame then repeats int32 codes instead of G·T Python string
+    objects (10x smaller, no object churn); values compare equal to the
+    original strings. Pass ``categorical_keys=False`` for plain object
+    columns.
+    """
     G, T = panel.shape
     data = {}
+    rep = np.repeat(np.arange(G, dtype=np.int32), T)
     if isinstance(gindex, pd.MultiIndex):
         for li, name in enumerate(keys):
-            data[name] = np.repeat(gindex.get_level_values(li).to_numpy(), T)
+            if categorical_keys:
+                data[name] = pd.Categorical.from_codes(
+                    np.asarray(gindex.codes[li], dtype=np.int32)[rep],
+                    categories=gindex.levels[li])
+            else:
+                data[name] = np.repeat(
+                    gindex.get_level_values(li).to_numpy(), T)
     else:
-        data[keys[0]] = np.repeat(np.asarray(gindex), T)
+        if categorical_keys:
+            data[keys[0]] = pd.Categorical.from_codes(
+                rep, categories=pd.Index(np.asarray(gindex)))
+        else:
+            data[keys[0]] = np.repeat(np.asarray(gindex), T)
     data[time_col] = np.tile(np.asarray(tvals), G)
     for name, mat in value_cols:
         data[name] = mat.reshape(-1)
