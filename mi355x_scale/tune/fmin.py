@@ -66,6 +66,25 @@ class Trials:
     def __len__(self):
         return len(self.trials)
 
+    # --- JSONL persistence (the trial log makes a search
+    # resumable; hyperopt's fmin(trials_save_file=...) contract) ---
+
+    def to_jsonl(self, path: str) -> None:
+        import json
+        tmp = f"{path}.tmp"
+        with open(tmp, "w") as f:
+            for t in self.trials:
+                f.write(json.dumps(t, default=float) + "\n")
+        os.replace(tmp, path)  # atomic: a crash never truncates the log
+
+    @classmethod
+    def from_jsonl(cls, path: str, **kwargs) -> "Trials":
+        import json
+        tr = cls(**kwargs)
+        with open(path) as f:
+            tr.trials = [json.loads(line) for line in f if line.strip()]
+        return tr
+
 
 def _default_parallelism() -> int:
     try:
@@ -143,17 +162,38 @@ def _make_rng(rstate) -> np.random.Generator:
 
 def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
          trials: Optional[Trials] = None, rstate=None,
-         verbose: bool = False, **_ignored) -> Dict:
-    """Minimize ``fn`` over ``space``; returns the best parameter dict."""
+         verbose: bool = False, trials_save_file: Optional[str] = None,
+         **_ignored) -> Dict:
+    """Minimize ``fn`` over ``space``; returns the best parameter dict.
+
+    ``max_evals`` counts TOTAL trials in ``trials`` (hyperopt semantics):
+    passing pre-populated trials — e.g. ``Trials.from_jsonl(...)`` or an
+    existing ``trials_save_file`` — resumes the search, with completed
+    trials seeding the TPE posterior. ``trials_save_file`` persists the
+    trial log (JSONL, atomic rewrite) after every trial.
+    """
     rng = _make_rng(rstate)
+    if (trials is None and trials_save_file
+            and os.path.exists(trials_save_file)):
+        trials = Trials.from_jsonl(trials_save_file)
     trials = trials if trials is not None else Trials()
     optimizer = algo() if algo is not None else TPE()
-    history: List[tuple] = []   # [(params, loss)] of OK trials
+    # seed the posterior from any completed trials (resume path)
+    history: List[tuple] = [
+        ({k: v[0] for k, v in t["misc"]["vals"].items()},
+         t["result"]["loss"])
+        for t in trials.trials
+        if t["result"].get("status") == STATUS_OK
+        and t["result"].get("loss") is not None
+    ]
+    start = len(trials)
 
     def _record(tid, params, result):
         trials.record(tid, params, result)
         if result.get("status") == STATUS_OK and result.get("loss") is not None:
             history.append((params, result["loss"]))
+        if trials_save_file:
+            trials.to_jsonl(trials_save_file)
         if verbose:
             print(f"[fmin] trial {tid}: loss={result.get('loss')} "
                   f"status={result.get('status')} params={params}")
@@ -167,7 +207,7 @@ def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
 
     par = getattr(trials, "parallelism", 1)
     if par <= 1:
-        for tid in range(max_evals):
+        for tid in range(start, max_evals):
             params = optimizer.propose(space, history, rng)
             result = _eval_trial(fn, bind_params(space, params))
             _record(tid, params, result)
@@ -187,8 +227,8 @@ def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
                                  initializer=_worker_init,
                                  initargs=(dq,)) as pool:
             pending = {}
-            issued = 0
-            done_n = 0
+            issued = start
+            done_n = start
             while done_n < max_evals:
                 while len(pending) < par and issued < max_evals:
                     params = optimizer.propose(space, history, rng)
