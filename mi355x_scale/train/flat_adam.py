@@ -113,6 +113,11 @@ class FlatAdam:
         }
 
     def load_state_dict(self, sd: dict) -> None:
+        if "flat_params" not in sd:
+            raise ValueError(
+                "checkpoint holds a torch.optim state dict, not FlatAdam's "
+                "flat-buffer format — resume graph-mode runs with "
+                "graph-mode checkpoints (or load only the model weights)")
         self.flat_params.copy_(sd["flat_params"])
         self.exp_avg.copy_(sd["exp_avg"])
         self.exp_avg_sq.copy_(sd["exp_avg_sq"])
