@@ -23,6 +23,50 @@ import torch
 import torch.distributed as dist
 
 
+def allreduce_flat(flat: torch.Tensor, world_size: int,
+                   state: Optional[Dict] = None) -> None:
+    """Sum-all-reduce one flat fp32 gradient buffer across ranks.
+
+    Strategy ladder:
+      * gloo (CPU tests): host-staged;
+      * world >= 4 (``MI355X_ALLREDUCE_BF16`` != "0"): bf16-compressed —
+        at 4+ ranks the flat fp32 all-reduce is the inter-step serial
+        cost, and halving the xGMI bytes costs two extra passes over the
+        buffer (~20 us each for 45 MB). Gradient averaging in bf16 is
+        standard DDP practice. Same compression on gloo so the
+        world-size-4 CPU test covers the semantics;
+      * ``MI355X_P2P_ALLREDUCE=1``: the hand-written direct-xGMI path
+        (7 concurrent link reads vs the ring's single-link bound) —
+        needs the buffer from p2p_allreduce.alloc_shared;
+      * otherwise: stock RCCL ring over xGMI.
+
+    ``state`` carries lazy staging buffers across calls.
+    """
+    import os
+    state = state if state is not None else {}
+    compress = (world_size >= 4
+                and os.environ.get("MI355X_ALLREDUCE_BF16", "1") == "1")
+    if dist.get_backend() == "gloo":
+        h = (flat.to(torch.bfloat16) if compress else flat).cpu()
+        dist.all_reduce(h)
+        flat.copy_(h)
+        return
+    if compress:
+        if "bf16" not in state:
+            state["bf16"] = torch.empty_like(flat, dtype=torch.bfloat16)
+        state["bf16"].copy_(flat)
+        dist.all_reduce(state["bf16"])
+        flat.copy_(state["bf16"])
+        return
+    if os.environ.get("MI355X_P2P_ALLREDUCE") == "1":
+        if "p2p" not in state:
+            from ..parallel.p2p_allreduce import P2PAllReduce
+            state["p2p"] = P2PAllReduce(flat)
+        state["p2p"].all_reduce_()
+        return
+    dist.all_reduce(flat)  # RCCL over xGMI
+
+
 class GraphedTrainStep:
     def __init__(self, model, optimizer, example_batch: Dict[str, torch.Tensor],
                  autocast_dtype: Optional[torch.dtype] = torch.bfloat16,
@@ -30,8 +74,7 @@ class GraphedTrainStep:
         self.model = model
         self.optimizer = optimizer
         self.world_size = world_size
-        self._p2p = None  # lazy MI355X_P2P_ALLREDUCE=1 path
-        self._bf16_buf = None  # lazy bf16 all-reduce staging buffer
+        self._ar_state: Dict = {}  # lazy all-reduce buffers (see allreduce_flat)
         self.device = next(model.parameters()).device
         assert self.device.type == "cuda", "graph capture needs a GPU"
         self.autocast_dtype = autocast_dtype
@@ -114,36 +157,7 @@ class GraphedTrainStep:
             self.g_opt = g
 
     def _allreduce_grads(self) -> None:
-        if dist.get_backend() == "gloo":
-            # gloo (CPU test path): stage the flat buffer through host
-            h = self.flat_grads.cpu()
-            dist.all_reduce(h)
-            self.flat_grads.copy_(h)
-            return
-        import os
-        if (self.world_size >= 4
-                and os.environ.get("MI355X_ALLREDUCE_BF16", "1") == "1"):
-            # At 4+ ranks the flat fp32 all-reduce is the inter-step
-            # serial cost; casting to bf16 halves the xGMI bytes for two
-            # extra 45 MB-pass casts (~20 us each). Gradient averaging in
-            # bf16 is standard DDP practice.
-            if self._bf16_buf is None:
-                self._bf16_buf = torch.empty_like(self.flat_grads,
-                                                  dtype=torch.bfloat16)
-            self._bf16_buf.copy_(self.flat_grads)
-            dist.all_reduce(self._bf16_buf)
-            self.flat_grads.copy_(self._bf16_buf)
-            return
-        if os.environ.get("MI355X_P2P_ALLREDUCE") == "1":
-            # hand-written direct xGMI path (7 concurrent link reads vs
-            # the ring's single-link bound); opt-in, needs the flat
-            # buffer from alloc_shared (FlatAdam handles that)
-            if self._p2p is None:
-                from ..parallel.p2p_allreduce import P2PAllReduce
-                self._p2p = P2PAllReduce(self.flat_grads)
-            self._p2p.all_reduce_()
-            return
-        dist.all_reduce(self.flat_grads)  # RCCL over xGMI
+        allreduce_flat(self.flat_grads, self.world_size, self._ar_state)
 
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
         for k, v in batch.items():
