@@ -110,10 +110,13 @@ def main() -> None:
     use_graph = use_cuda and not args.no_graph
     runner = model
     if use_graph:
-        # fused flat Adam: one HIP kernel per optimizer step, and its
-        # flat grad buffer doubles as the single all-reduce target
+        # fused flat Adam: one HIP kernel per optimizer step, its flat
+        # grad buffers double as the all-reduce targets; bf16 working
+        # params (fp32 master) remove the per-step autocast weight casts
+        # and halve the gradient all-reduce bytes
         from mi355x_scale.train.flat_adam import FlatAdam
-        optimizer = FlatAdam(model.parameters(), lr=1e-5)
+        optimizer = FlatAdam(model.parameters(), lr=1e-5,
+                             bf16_params=use_cuda)
     else:
         if n_gpus > 1:
             kwargs = dict(bucket_cap_mb=args.bucket_cap_mb,

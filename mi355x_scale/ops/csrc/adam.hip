@@ -15,6 +15,7 @@
 // hipGraphs with no host-side state.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 __global__ void adam_bump_step_kernel(int* step) {
   if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
@@ -80,4 +81,58 @@ extern "C" void launch_adam_step(float* p, const float* g, float* m,
   hipLaunchKernelGGL(adam_step_kernel, dim3(grid), dim3(block), 0, stream,
                      p, g, m, v, step, lr, beta1, beta2, eps, weight_decay,
                      n);
+}
+
+// ------------------------------------------------------------- mixed dtype
+// bf16-parameter mode: elements [0, nb) have bf16 gradients (gb) and a
+// bf16 working copy of the parameter (pb, rewritten from the fp32
+// master every step — the same rounding autocast's per-step weight cast
+// performed, minus ~40 cast kernels per replay); elements [nb, n) are
+// fp32 params (BN scale/bias) with fp32 gradients (gf).
+__global__ __launch_bounds__(256) void adam_step_mixed_kernel(
+    float* __restrict__ master, const __hip_bfloat16* __restrict__ gb,
+    const float* __restrict__ gf, float* __restrict__ m,
+    float* __restrict__ v, __hip_bfloat16* __restrict__ pb,
+    const int* __restrict__ step, float lr, float beta1, float beta2,
+    float eps, float weight_decay, long long nb, long long n) {
+  const int t = *step;
+  const float bc1 = 1.f - powf(beta1, (float)t);
+  const float bc2 = 1.f - powf(beta2, (float)t);
+  const float step_size = lr / bc1;
+  const long long stride = (long long)gridDim.x * blockDim.x * 4;
+
+  for (long long base = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       base < n; base += stride) {
+    const int lanes = (int)min((long long)4, n - base);
+    for (int j = 0; j < lanes; ++j) {
+      const long long i = base + j;
+      float g = i < nb ? __bfloat162float(gb[i]) : gf[i - nb];
+      float p = master[i];
+      g += weight_decay * p;
+      float mj = beta1 * m[i] + (1.f - beta1) * g;
+      float vj = beta2 * v[i] + (1.f - beta2) * g * g;
+      m[i] = mj;
+      v[i] = vj;
+      p -= step_size * mj / (sqrtf(vj / bc2) + eps);
+      master[i] = p;
+      if (i < nb) pb[i] = __float2bfloat16(p);
+    }
+  }
+}
+
+extern "C" void launch_adam_step_mixed(float* master, const void* gb,
+                                       const float* gf, float* m, float* v,
+                                       void* pb, int* step, float lr,
+                                       float beta1, float beta2, float eps,
+                                       float weight_decay, long long nb,
+                                       long long n, hipStream_t stream) {
+  hipLaunchKernelGGL(adam_bump_step_kernel, dim3(1), dim3(64), 0, stream,
+                     step);
+  const int block = 256;
+  long long want = (n + block * 4 - 1) / (block * 4);
+  int grid = (int)(want < 1 ? 1 : (want > 2080 ? 2080 : want));
+  hipLaunchKernelGGL(adam_step_mixed_kernel, dim3(grid), dim3(block), 0,
+                     stream, master, (const __hip_bfloat16*)gb, gf, m, v,
+                     (__hip_bfloat16*)pb, step, lr, beta1, beta2, eps,
+                     weight_decay, nb, n);
 }

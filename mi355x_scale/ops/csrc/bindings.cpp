@@ -384,6 +384,39 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                    at::cuda::getCurrentHIPStream().stream());
 }
 
+extern "C" void launch_adam_step_mixed(float* master, const void* gb,
+                                       const float* gf, float* m, float* v,
+                                       void* pb, int* step, float lr,
+                                       float beta1, float beta2, float eps,
+                                       float weight_decay, long long nb,
+                                       long long n, hipStream_t stream);
+
+// bf16-parameter FlatAdam step: master fp32 [n]; gb bf16 grads for the
+// first nb elements (pb = bf16 working params, rewritten here); gf fp32
+// grads for the rest.
+void adam_step_mixed(torch::Tensor master, torch::Tensor gb,
+                     torch::Tensor gf, torch::Tensor m, torch::Tensor v,
+                     torch::Tensor pb, torch::Tensor step, double lr,
+                     double beta1, double beta2, double eps,
+                     double weight_decay) {
+  _check_f32(master, "master"); _check_f32(m, "m"); _check_f32(v, "v");
+  TORCH_CHECK(gb.scalar_type() == torch::kBFloat16 &&
+              pb.scalar_type() == torch::kBFloat16, "gb/pb must be bf16");
+  TORCH_CHECK(step.is_cuda() && step.scalar_type() == torch::kInt32 &&
+              step.numel() == 1, "step must be int32[1] on device");
+  long long n = master.numel();
+  long long nb = gb.numel();
+  TORCH_CHECK(pb.numel() == nb && gf.numel() == n - nb &&
+              m.numel() == n && v.numel() == n, "buffer length mismatch");
+  const float* gf_p = gf.numel() ? gf.data_ptr<float>() : nullptr;
+  launch_adam_step_mixed(master.data_ptr<float>(), gb.data_ptr(), gf_p,
+                         m.data_ptr<float>(), v.data_ptr<float>(),
+                         pb.data_ptr(), step.data_ptr<int>(), (float)lr,
+                         (float)beta1, (float)beta2, (float)eps,
+                         (float)weight_decay, nb, n,
+                         at::cuda::getCurrentHIPStream().stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
@@ -406,4 +439,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flat Adam step (one kernel over p/g/m/v)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
+  m.def("adam_step_mixed", &adam_step_mixed,
+        "fused flat Adam with bf16 params/grads + fp32 master");
 }

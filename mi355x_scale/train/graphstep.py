@@ -44,12 +44,15 @@ def allreduce_flat(flat: torch.Tensor, world_size: int,
     """
     import os
     state = state if state is not None else {}
-    compress = (world_size >= 4
+    compress = (world_size >= 4 and flat.dtype == torch.float32
                 and os.environ.get("MI355X_ALLREDUCE_BF16", "1") == "1")
     if dist.get_backend() == "gloo":
         h = (flat.to(torch.bfloat16) if compress else flat).cpu()
         dist.all_reduce(h)
         flat.copy_(h)
+        return
+    if flat.dtype == torch.bfloat16:
+        dist.all_reduce(flat)  # already half-width (bf16-param grads)
         return
     if compress:
         if "bf16" not in state:
@@ -58,7 +61,8 @@ def allreduce_flat(flat: torch.Tensor, world_size: int,
         dist.all_reduce(state["bf16"])
         flat.copy_(state["bf16"])
         return
-    if os.environ.get("MI355X_P2P_ALLREDUCE") == "1":
+    if (os.environ.get("MI355X_P2P_ALLREDUCE") == "1"
+            and flat.dtype == torch.float32):
         if "p2p" not in state:
             from ..parallel.p2p_allreduce import P2PAllReduce
             state["p2p"] = P2PAllReduce(flat)
@@ -74,7 +78,6 @@ class GraphedTrainStep:
         self.model = model
         self.optimizer = optimizer
         self.world_size = world_size
-        self._ar_state: Dict = {}  # lazy all-reduce buffers (see allreduce_flat)
         self.device = next(model.parameters()).device
         assert self.device.type == "cuda", "graph capture needs a GPU"
         self.autocast_dtype = autocast_dtype
@@ -84,17 +87,21 @@ class GraphedTrainStep:
             k: v.to(self.device).clone() for k, v in example_batch.items()
         }
 
-        # Flat gradient buffer; every p.grad is a view into it so the
-        # cross-rank all-reduce is ONE call. FlatAdam (train/flat_adam.py)
-        # already owns such a buffer and has the views installed — reuse
-        # it so optimizer + all-reduce share one layout.
-        if hasattr(optimizer, "flat_grads"):
-            self.flat_grads = optimizer.flat_grads
+        # Flat gradient buffer(s); every p.grad is a view into one so the
+        # cross-rank all-reduce is one call per buffer. FlatAdam
+        # (train/flat_adam.py) already owns them with the views installed
+        # — reuse so optimizer + all-reduce share one layout (bf16-param
+        # mode has two: a bf16 buffer for matrix params, fp32 for the
+        # rest).
+        if hasattr(optimizer, "grad_buffers"):
+            self.grad_buffers = list(optimizer.grad_buffers)
+        elif hasattr(optimizer, "flat_grads"):
+            self.grad_buffers = [optimizer.flat_grads]
         else:
             params = [p for p in model.parameters() if p.requires_grad]
             total = sum(p.numel() for p in params)
-            self.flat_grads = torch.zeros(total, device=self.device,
-                                          dtype=torch.float32)
+            flat = torch.zeros(total, device=self.device,
+                               dtype=torch.float32)
             off = 0
             for p in params:
                 # grad views adopt each param's own (dense) stride order —
@@ -103,12 +110,16 @@ class GraphedTrainStep:
                 # permute (the "gradient layout contract") and the flat
                 # all-reduce still sums identical byte layouts across
                 # ranks.
-                p.grad = self.flat_grads[off:off + p.numel()].as_strided(
+                p.grad = flat[off:off + p.numel()].as_strided(
                     p.shape, p.stride())
                 off += p.numel()
+            self.grad_buffers = [flat]
+        self.flat_grads = self.grad_buffers[0]  # back-compat alias
+        self._ar_states = [dict() for _ in self.grad_buffers]
 
         def _fwd_bwd():
-            self.flat_grads.zero_()
+            for b in self.grad_buffers:
+                b.zero_()
             with torch.autocast(device_type="cuda", dtype=autocast_dtype,
                                 enabled=autocast_dtype is not None):
                 loss = model.training_step(self.static_batch, 0)
@@ -125,7 +136,8 @@ class GraphedTrainStep:
                     # identical to the replay-time step: keep ranks'
                     # parameters bit-identical through warmup too
                     self._allreduce_grads()
-                    self.flat_grads.div_(world_size)
+                    for b in self.grad_buffers:
+                        b.div_(world_size)
                 optimizer.step()
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
@@ -152,12 +164,14 @@ class GraphedTrainStep:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 if world_size > 1:
-                    self.flat_grads.div_(world_size)
+                    for b in self.grad_buffers:
+                        b.div_(world_size)
                 optimizer.step()
             self.g_opt = g
 
     def _allreduce_grads(self) -> None:
-        allreduce_flat(self.flat_grads, self.world_size, self._ar_state)
+        for b, st in zip(self.grad_buffers, self._ar_states):
+            allreduce_flat(b, self.world_size, st)
 
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
         for k, v in batch.items():
@@ -169,6 +183,7 @@ class GraphedTrainStep:
             self.g_opt.replay()
         else:
             if self.world_size > 1:
-                self.flat_grads.div_(self.world_size)
+                for b in self.grad_buffers:
+                    b.div_(self.world_size)
             self.optimizer.step()
         return self.static_loss

@@ -126,7 +126,8 @@ class Trainer:
             # buffer is the graphed step's single all-reduce target
             from .flat_adam import FlatAdam
             optimizer = FlatAdam(model.parameters(),
-                                 lr=getattr(model, "lr", 1e-3))
+                                 lr=getattr(model, "lr", 1e-3),
+                                 bf16_params=device.type == "cuda")
         else:
             optimizer = model.configure_optimizers()
         start_epoch = 0

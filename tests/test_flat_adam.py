@@ -115,3 +115,64 @@ def test_gpu_graph_capture_replay():
     # on-device counter
     assert int(opt.step_t.item()) == t0 + 5
     assert torch.isfinite(opt.flat_params).all()
+
+
+def _conv_model(seed=0):
+    torch.manual_seed(seed)
+    return nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.BatchNorm2d(8),
+                         nn.ReLU(), nn.Flatten(), nn.Linear(8 * 16, 4))
+
+
+def test_bf16_params_cpu_semantics():
+    """bf16-param mode: matrix params become bf16 views of the flat bf16
+    buffer, 1-D params stay fp32 master views; the bf16 copies always
+    equal the rounded master."""
+    m = _conv_model()
+    opt = FlatAdam(m.parameters(), lr=1e-2, bf16_params=True)
+    conv_w = m[0].weight
+    bn_w = m[1].weight
+    assert conv_w.dtype == torch.bfloat16
+    assert bn_w.dtype == torch.float32
+    g = torch.Generator().manual_seed(3)
+    for _ in range(4):
+        opt.zero_grad()
+        x = torch.randn(2, 3, 4, 4, generator=g)
+        # CPU conv wants uniform dtype: run the model in fp32 shadow by
+        # exercising only the optimizer contract here
+        for p in opt.params:
+            p.grad.copy_(torch.randn(p.shape, generator=g).to(p.grad.dtype))
+        opt.step()
+    assert int(opt.step_t.item()) == 4
+    torch.testing.assert_close(
+        opt.flat_pb16, opt.flat_master[:opt.n_bf16].bfloat16())
+    assert torch.isfinite(opt.flat_master).all()
+
+
+def test_bf16_params_master_follows_adam_math():
+    """The mixed-mode fp32 master must follow the exact Adam recurrence
+    given the (bf16-rounded) gradients actually applied."""
+    m = _conv_model()
+    opt = FlatAdam(m.parameters(), lr=1e-2, bf16_params=True)
+    master0 = opt.flat_master.clone()
+    exp_m = torch.zeros_like(master0)
+    exp_v = torch.zeros_like(master0)
+    expect = master0.clone()
+    g = torch.Generator().manual_seed(5)
+    for t in range(1, 6):
+        opt.zero_grad()
+        flat_g = torch.empty_like(master0)
+        off = 0
+        for p in opt.params:
+            gr = torch.randn(p.shape, generator=g).to(p.grad.dtype)
+            p.grad.copy_(gr)
+            flat_g[off:off + p.numel()].as_strided(
+                p.shape, p.stride()).copy_(gr.float())
+            off += p.numel()
+        opt.step()
+        exp_m.mul_(0.9).add_(flat_g, alpha=0.1)
+        exp_v.mul_(0.999).addcmul_(flat_g, flat_g, value=0.001)
+        bc1, bc2 = 1 - 0.9 ** t, 1 - 0.999 ** t
+        expect.addcdiv_(exp_m, (exp_v / bc2).sqrt().add_(1e-8),
+                        value=-1e-2 / bc1)
+    torch.testing.assert_close(opt.flat_master, expect,
+                               atol=1e-6, rtol=1e-5)
