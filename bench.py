@@ -96,7 +96,11 @@ def main() -> None:
     data_dir = None
     if not args.resident:
         rpg = batch if use_cuda else 16
-        rows = max(args.rows, 20 * rpg * n_gpus)
+        # size the dataset for the LARGEST world (8) regardless of N so
+        # the driver's N=1,2,4,8 sequence generates the parquet once and
+        # reuses it (the reader is infinite; extra rows just mean more
+        # distinct batches per shard)
+        rows = max(args.rows, 20 * rpg * (8 if use_cuda else n_gpus))
         rows = (rows // rpg) * rpg
         data_dir = prepare_data(rows, image_hw, ctx.rank, n_gpus, rpg=rpg)
 
