@@ -142,3 +142,38 @@ def test_mfma_path_equals_lane_path():
     import torch
     rel = ((m1 - m2).abs() / (m2.abs() + 1e-3)).max().item()
     assert rel < 1e-3, rel
+
+
+@pytest.mark.gpu
+def test_gpu_pipeline_dataframe_to_forecast_frame():
+    """Whole W1 job on GPU: long DataFrame (arrow-backed keys) ->
+    C++ group gather -> batched CDNA4 fit -> long forecast frame
+    (run_fine_grained_forecast_gpu; loads groupby._gather + ops._C)."""
+    import pandas as pd
+
+    from mi355x_scale.forecast.pipeline import run_fine_grained_forecast_gpu
+    from mi355x_scale.groupby.gather import HAVE_GATHER_EXT
+
+    assert HAVE_GATHER_EXT, "C++ gather engine must be built on GPU boxes"
+    rng = np.random.default_rng(123)
+    G, T = 500, 157
+    dates = pd.date_range("2020-06-29", periods=T, freq="W-MON")
+    base = rng.uniform(400, 1200, size=(G, 1))
+    y = base + rng.normal(0, base * 0.05, size=(G, T))
+    df = pd.DataFrame({
+        "Product": pd.array(np.repeat([f"P{i % 7}" for i in range(G)], T),
+                            dtype="string[pyarrow]"),
+        "SKU": pd.array(np.repeat([f"S{i:05d}" for i in range(G)], T),
+                        dtype="string[pyarrow]"),
+        "Date": np.tile(dates.to_numpy(), G),
+        "Demand": np.clip(y, 0, None).reshape(-1),
+    })
+    res = run_fine_grained_forecast_gpu(df)
+    assert len(res) == G * T
+    assert list(res.columns) == ["Product", "SKU", "Date", "Demand",
+                                 "Demand_Fitted"]
+    assert res["Demand_Fitted"].notna().mean() > 0.95
+    # fitted values must track the actual demand scale per group
+    one = res[res["SKU"] == "S00007"]
+    ratio = one["Demand_Fitted"].mean() / max(one["Demand"].mean(), 1.0)
+    assert 0.5 < ratio < 1.5, ratio
