@@ -91,8 +91,17 @@ __global__ __launch_bounds__(256) void maxpool_fwd_kernel(
   }
 }
 
-// dx[n,h,w,c] = sum over output windows (ho,wo) containing (h,w) of
-// dy[n,ho,wo,c] where code matches this input position.
+// Backward, one 2x2 INPUT tile per thread-lane. The tile at input
+// origin (2i, 2j) is touched only by the four output windows
+// (i,j), (i,j+1), (i+1,j), (i+1,j+1), and each of the tile's pixels sits
+// at a FIXED position inside those windows (k=3, s=2, p=1):
+//     (2i  ,2j  ): (i,j)@4
+//     (2i  ,2j+1): (i,j)@5   (i,j+1)@3
+//     (2i+1,2j  ): (i,j)@7   (i+1,j)@1
+//     (2i+1,2j+1): (i,j)@8   (i,j+1)@6  (i+1,j)@2  (i+1,j+1)@0
+// so every dy/code vector is loaded once and reused for the whole tile
+// (2.25x fewer loads than the per-pixel form, and all code compares are
+// against constants). Tiles == the output grid: i in [0,Ho), j in [0,Wo).
 __global__ __launch_bounds__(256) void maxpool_bwd_kernel(
     const bf16* __restrict__ dy, const unsigned char* __restrict__ code,
     bf16* __restrict__ dx, int N, int H, int W, int Ho, int Wo, int C) {
@@ -100,53 +109,67 @@ __global__ __launch_bounds__(256) void maxpool_bwd_kernel(
   const int lane = threadIdx.x % lanes;
   const int rsub = threadIdx.x / lanes;
   const int rows_per_iter = blockDim.x / lanes;
-  const long long in_rows = (long long)N * H * W;
+  const long long tiles = (long long)N * Ho * Wo;
   const long long rstride = (long long)gridDim.x * rows_per_iter;
 
   for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
-       row < in_rows; row += rstride) {
-    int w = (int)(row % W);
-    int h = (int)((row / W) % H);
-    int n = (int)(row / ((long long)H * W));
+       row < tiles; row += rstride) {
+    int j = (int)(row % Wo);
+    int i = (int)((row / Wo) % Ho);
+    int n = (int)(row / ((long long)Ho * Wo));
+    const long long lvec = (long long)lane * VEC;
 
-    float acc[VEC];
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
-
-    // windows: ho with 2*ho-1 <= h <= 2*ho+1  =>  ceil((h-1)/2) <= ho
-    // <= floor((h+1)/2); h >= 0 so ceil((h-1)/2) == h/2
-    int ho_lo = h >> 1;
-    int ho_hi = (h + 1) >> 1;
-    int wo_lo = w >> 1;
-    int wo_hi = (w + 1) >> 1;
-    if (ho_lo < 0) ho_lo = 0;
-    if (wo_lo < 0) wo_lo = 0;
-    if (ho_hi >= Ho) ho_hi = Ho - 1;
-    if (wo_hi >= Wo) wo_hi = Wo - 1;
-
-    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
-      int ih = h - (2 * ho - 1);
-      if (ih < 0 || ih > 2) continue;
-      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
-        int iw = w - (2 * wo - 1);
-        if (iw < 0 || iw > 2) continue;
-        unsigned char want = (unsigned char)(ih * 3 + iw);
-        const long long ooff =
-            (((long long)n * Ho + ho) * Wo + wo) * C + (long long)lane * VEC;
-        CVec cv;
-        cv.u = *reinterpret_cast<const uint2*>(code + ooff);
-        BVec g;
-        g.u = *reinterpret_cast<const uint4*>(dy + ooff);
-#pragma unroll
-        for (int j = 0; j < VEC; ++j)
-          if (cv.c[j] == want) acc[j] += __bfloat162float(g.h[j]);
+    // load the (up to) four windows' dy + code vectors
+    BVec g00{}, g01{}, g10{}, g11{};
+    CVec c00{}, c01{}, c10{}, c11{};
+    bool v01 = (j + 1 < Wo), v10 = (i + 1 < Ho);
+    {
+      const long long base =
+          (((long long)n * Ho + i) * Wo + j) * C + lvec;
+      g00.u = *reinterpret_cast<const uint4*>(dy + base);
+      c00.u = *reinterpret_cast<const uint2*>(code + base);
+      if (v01) {
+        g01.u = *reinterpret_cast<const uint4*>(dy + base + C);
+        c01.u = *reinterpret_cast<const uint2*>(code + base + C);
+      }
+      if (v10) {
+        const long long b2 = base + (long long)Wo * C;
+        g10.u = *reinterpret_cast<const uint4*>(dy + b2);
+        c10.u = *reinterpret_cast<const uint2*>(code + b2);
+        if (v01) {
+          g11.u = *reinterpret_cast<const uint4*>(dy + b2 + C);
+          c11.u = *reinterpret_cast<const uint2*>(code + b2 + C);
+        }
       }
     }
-    const long long ioff = row * C + (long long)lane * VEC;
-    BVec o;
+
+    const int h0 = 2 * i, w0 = 2 * j;
+    const bool r1 = (h0 + 1 < H), cL1 = (w0 + 1 < W);
+    BVec o00, o01, o10, o11;
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) o.h[j] = __float2bfloat16(acc[j]);
-    *reinterpret_cast<uint4*>(dx + ioff) = o.u;
+    for (int k = 0; k < VEC; ++k) {
+      float a00 = (c00.c[k] == 4) ? __bfloat162float(g00.h[k]) : 0.f;
+      float a01 = (c00.c[k] == 5) ? __bfloat162float(g00.h[k]) : 0.f;
+      if (v01 && c01.c[k] == 3) a01 += __bfloat162float(g01.h[k]);
+      float a10 = (c00.c[k] == 7) ? __bfloat162float(g00.h[k]) : 0.f;
+      if (v10 && c10.c[k] == 1) a10 += __bfloat162float(g10.h[k]);
+      float a11 = (c00.c[k] == 8) ? __bfloat162float(g00.h[k]) : 0.f;
+      if (v01 && c01.c[k] == 6) a11 += __bfloat162float(g01.h[k]);
+      if (v10 && c10.c[k] == 2) a11 += __bfloat162float(g10.h[k]);
+      if (v10 && v01 && c11.c[k] == 0) a11 += __bfloat162float(g11.h[k]);
+      o00.h[k] = __float2bfloat16(a00);
+      o01.h[k] = __float2bfloat16(a01);
+      o10.h[k] = __float2bfloat16(a10);
+      o11.h[k] = __float2bfloat16(a11);
+    }
+    const long long ibase = (((long long)n * H + h0) * W + w0) * C + lvec;
+    *reinterpret_cast<uint4*>(dx + ibase) = o00.u;
+    if (cL1) *reinterpret_cast<uint4*>(dx + ibase + C) = o01.u;
+    if (r1) {
+      const long long ib2 = ibase + (long long)W * C;
+      *reinterpret_cast<uint4*>(dx + ib2) = o10.u;
+      if (cL1) *reinterpret_cast<uint4*>(dx + ib2 + C) = o11.u;
+    }
   }
 }
 
