@@ -172,7 +172,8 @@ extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* invstd,
                                        const float* weight, float* dweight,
                                        float* dbias, float* k, long long rows,
-                                       int C, hipStream_t stream);
+                                       int C, int accumulate,
+                                       hipStream_t stream);
 extern "C" void launch_bn_bwd_apply(const void* dz, const void* y,
                                     const void* x, const float* mean,
                                     const float* invstd, const float* k,
@@ -286,7 +287,7 @@ void bn_bwd_apply_dym(torch::Tensor dym, torch::Tensor x,
 void bn_bwd_finalize(torch::Tensor partial, torch::Tensor invstd,
                      torch::Tensor weight, torch::Tensor dweight,
                      torch::Tensor dbias, torch::Tensor k, int64_t rows,
-                     int64_t C) {
+                     int64_t C, bool accumulate) {
   _check_f32(partial, "partial"); _check_f32(invstd, "invstd");
   _check_f32(weight, "weight"); _check_f32(dweight, "dweight");
   _check_f32(dbias, "dbias"); _check_f32(k, "k");
@@ -297,7 +298,8 @@ void bn_bwd_finalize(torch::Tensor partial, torch::Tensor invstd,
                          invstd.data_ptr<float>(),
                          weight.data_ptr<float>(), dweight.data_ptr<float>(),
                          dbias.data_ptr<float>(), k.data_ptr<float>(), rows,
-                         (int)C, at::cuda::getCurrentHIPStream().stream());
+                         (int)C, accumulate ? 1 : 0,
+                         at::cuda::getCurrentHIPStream().stream());
 }
 
 void bn_bwd_apply(torch::Tensor dz, torch::Tensor y, torch::Tensor x,

@@ -275,7 +275,7 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, int nblocks,
     const float* __restrict__ invstd, const float* __restrict__ weight,
     float* __restrict__ dweight, float* __restrict__ dbias,
-    float* __restrict__ k, long long rows, int C) {
+    float* __restrict__ k, long long rows, int C, int accumulate) {
   int c = blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
   int lane = threadIdx.x % 64;
   if (c >= C) return;
@@ -290,8 +290,10 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     sdyx += __shfl_down(sdyx, off, 64);
   }
   if (lane != 0) return;
-  dbias[c] = sdy;
-  dweight[c] = sdyx;
+  // accumulate=1: write straight into the flat grad views (graph-capture
+  // path) so autograd's per-param AccumulateGrad add kernels disappear
+  dbias[c] = accumulate ? dbias[c] + sdy : sdy;
+  dweight[c] = accumulate ? dweight[c] + sdyx : sdyx;
   k[c] = weight[c] * invstd[c];  // the dx scale factor
   k[C + c] = sdy / (float)rows;
   k[2 * C + c] = sdyx / (float)rows;
@@ -475,10 +477,11 @@ extern "C" void launch_bn_bwd_finalize(const float* partial, int nblocks,
                                        const float* invstd,
                                        const float* weight, float* dweight,
                                        float* dbias, float* k, long long rows,
-                                       int C, hipStream_t stream) {
+                                       int C, int accumulate,
+                                       hipStream_t stream) {
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 3) / 4),
                      dim3(256), 0, stream, partial, nblocks, invstd, weight,
-                     dweight, dbias, k, rows, C);
+                     dweight, dbias, k, rows, C, accumulate);
 }
 
 extern "C" void launch_bn_bwd_apply(const void* dz, const void* y,

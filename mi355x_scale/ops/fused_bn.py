@@ -52,22 +52,42 @@ class _FusedBNFn(torch.autograd.Function):
         y = torch.empty_like(x)
         res = residual if residual is not None else x.new_empty(0)
         _C.bn_fwd_apply(x, res, y, mean, invstd, weight, bias, C, relu)
-        ctx.save_for_backward(x, y, mean, invstd, weight)
+        ctx.save_for_backward(x, y, mean, invstd, weight, bias)
         ctx.relu = relu
         ctx.has_res = residual is not None
         return y
 
     @staticmethod
     def backward(ctx, dz):
-        x, y, mean, invstd, weight = ctx.saved_tensors
+        x, y, mean, invstd, weight, bias = ctx.saved_tensors
         C = x.shape[1]
         rows = x.numel() // C
         dev = x.device
         if not dz.is_contiguous(memory_format=torch.channels_last):
             dz = dz.contiguous(memory_format=torch.channels_last)
         k = torch.empty(3 * C, dtype=torch.float32, device=dev)
-        dweight = torch.empty(C, dtype=torch.float32, device=dev)
-        dbias = torch.empty(C, dtype=torch.float32, device=dev)
+        # Inside hipGraph capture (the flagship graphed step), write the
+        # channel grads straight into the existing flat grad views and
+        # return None for weight/bias — autograd's per-param
+        # AccumulateGrad adds (~40 launch-bound kernels/step across the
+        # model's BNs) vanish from the captured graph. Outside capture
+        # (eager, DDP hooks, tests with fresh .grad) keep the standard
+        # return-grads contract.
+        def _grad_view(p):
+            g = p.grad
+            if (g is not None and g.is_cuda and g.is_contiguous()
+                    and g.dtype == torch.float32 and g.numel() == C):
+                return g
+            return None
+
+        wg, bg = _grad_view(weight), _grad_view(bias)
+        fuse_acc = (torch.cuda.is_current_stream_capturing()
+                    and wg is not None and bg is not None)
+        if fuse_acc:
+            dweight, dbias = wg, bg
+        else:
+            dweight = torch.empty(C, dtype=torch.float32, device=dev)
+            dbias = torch.empty(C, dtype=torch.float32, device=dev)
         dx = torch.empty_like(x)
         if ctx.has_res:
             # residual layer: the reduce pass materializes the masked
@@ -78,17 +98,19 @@ class _FusedBNFn(torch.autograd.Function):
             partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, dym, C,
                                        ctx.relu)
             _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
-                               rows, C)
+                               rows, C, fuse_acc)
             _C.bn_bwd_apply_dym(dym, x, mean, invstd, k, dx, C)
-            return (dx, dym, dweight, dbias,
-                    None, None, None, None, None)
+            return (dx, dym, None if fuse_acc else dweight,
+                    None if fuse_acc else dbias, None, None, None, None,
+                    None)
         partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, x.new_empty(0),
                                    C, ctx.relu)
         _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
-                           rows, C)
+                           rows, C, fuse_acc)
         _C.bn_bwd_apply(dz, y, x, mean, invstd, k, dx, x.new_empty(0), C,
                         ctx.relu)
-        return (dx, None, dweight, dbias, None, None, None, None, None)
+        return (dx, None, None if fuse_acc else dweight,
+                None if fuse_acc else dbias, None, None, None, None, None)
 
 
 class FusedBNReLU2d(nn.Module):
