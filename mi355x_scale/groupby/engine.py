@@ -24,9 +24,8 @@ from __future__ import annotations
 
 import traceback
 from concurrent.futures import ProcessPoolExecutor
-from typing import Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Callable, List, Optional, Sequence, Tuple
 
-import numpy as np
 import pandas as pd
 
 _TYPE_MAP = {

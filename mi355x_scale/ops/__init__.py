@@ -8,7 +8,6 @@ implementations run (they are also the numerics references in tests).
 """
 from __future__ import annotations
 
-import torch
 
 try:
     from . import _C  # type: ignore

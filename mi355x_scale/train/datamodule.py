@@ -19,7 +19,7 @@ import numpy as np
 import torch
 
 from ..data import (BatchReader, DataLoader, DatasetManifest, DeviceLoader,
-                    TransformSpec)
+                    TransformSpec)  # noqa: F401 (re-export)
 from .module import DataModule
 
 

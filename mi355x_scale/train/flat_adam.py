@@ -27,7 +27,6 @@ the flat buffers — that path is the numerics oracle in the tests.
 """
 from __future__ import annotations
 
-import math
 from typing import Iterable, List, Tuple
 
 import torch

@@ -24,7 +24,7 @@ from typing import Any, Callable, Dict, List, Optional
 import numpy as np
 
 from .space import bind_params
-from .tpe import TPE, _RandomAlgo
+from .tpe import TPE
 
 STATUS_OK = "ok"
 STATUS_FAIL = "fail"
