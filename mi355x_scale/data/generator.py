@@ -49,6 +49,36 @@ def arma_generate_sample(ar, ma, nsample: int, scale: float = 1.0,
     return x[burnin:]
 
 
+def arma_generate_sample_batch_gpu(ar, ma, nsample: int,
+                                   scale=1.0, burnin: int = 3000,
+                                   rng: Optional[np.random.Generator] = None,
+                                   device="cuda:0"):
+    """Batched GPU form of :func:`arma_generate_sample`:
+    one group per lane, time-major [T][G] output.
+
+    ``ar``/``ma``: [G, na]/[G, nb] float arrays with the statsmodels
+    leading-1 convention; ``scale``: scalar or [G]. Returns a torch
+    tensor [nsample, G] on ``device``.
+    """
+    import torch
+
+    from ..ops import _C, require_ext
+    require_ext()
+    ar_t = torch.as_tensor(np.asarray(ar, dtype=np.float32),
+                           device=device).contiguous()
+    ma_t = torch.as_tensor(np.asarray(ma, dtype=np.float32),
+                           device=device).contiguous()
+    G = ar_t.shape[0]
+    rng = rng or np.random.default_rng(SEED)
+    eps = rng.standard_normal((nsample + burnin, G)).astype(np.float32)
+    eps = eps * np.broadcast_to(np.asarray(scale, dtype=np.float32),
+                                (G,))[None, :]
+    eps_t = torch.as_tensor(np.ascontiguousarray(eps), device=device)
+    out = torch.empty((nsample, G), dtype=torch.float32, device=device)
+    _C.arma_generate(ar_t, ma_t, eps_t, out, burnin)
+    return out
+
+
 def _week_dates(n_weeks: int, end: str = "2023-06-26") -> pd.DatetimeIndex:
     endts = pd.Timestamp(end)
     return pd.date_range(end=endts, periods=n_weeks, freq="W-MON")

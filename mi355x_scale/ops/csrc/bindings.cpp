@@ -322,6 +322,32 @@ void bn_bwd_apply(torch::Tensor dz, torch::Tensor y, torch::Tensor x,
                       at::cuda::getCurrentHIPStream().stream());
 }
 
+// ---------------------------------------------------------------- arma gen
+extern "C" void launch_arma_gen(const float* ar, const float* ma,
+                                const float* eps, float* out, int na,
+                                int nb, int burn, int T, long long G,
+                                hipStream_t stream);
+
+// Batched ARMA sample paths: ar [G][na] (leading 1), ma [G][nb],
+// eps [burn+T][G] time-major noise, out [T][G].
+void arma_generate(torch::Tensor ar, torch::Tensor ma, torch::Tensor eps,
+                   torch::Tensor out, int64_t burn) {
+  _check_f32(ar, "ar"); _check_f32(ma, "ma"); _check_f32(eps, "eps");
+  _check_f32(out, "out");
+  long long G = ar.size(0);
+  int na = (int)ar.size(1), nb = (int)ma.size(1);
+  TORCH_CHECK(na >= 1 && na <= 5 && nb >= 1 && nb <= 5,
+              "orders up to 4 (leading-1 length 5)");
+  TORCH_CHECK(ma.size(0) == G && out.size(1) == G && eps.size(1) == G,
+              "G mismatch");
+  int T = (int)out.size(0);
+  TORCH_CHECK(eps.size(0) == burn + T, "eps must be [burn+T][G]");
+  launch_arma_gen(ar.data_ptr<float>(), ma.data_ptr<float>(),
+                  eps.data_ptr<float>(), out.data_ptr<float>(), na, nb,
+                  (int)burn, T, G,
+                  at::cuda::getCurrentHIPStream().stream());
+}
+
 // ----------------------------------------------------------------- maxpool
 extern "C" void launch_maxpool_fwd(const void* x, void* y,
                                    unsigned char* code, int N, int H, int W,
@@ -443,4 +469,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("adam_step_mixed", &adam_step_mixed,
         "fused flat Adam with bf16 params/grads + fp32 master");
+  m.def("arma_generate", &arma_generate,
+        "batched ARMA(p,q) sample paths, one group per lane");
 }

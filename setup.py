@@ -21,6 +21,7 @@ SRC = [
     "mi355x_scale/ops/csrc/fused_bn.hip",
     "mi355x_scale/ops/csrc/adam.hip",
     "mi355x_scale/ops/csrc/maxpool.hip",
+    "mi355x_scale/ops/csrc/arma_gen.hip",
 ]
 
 setup(
