@@ -51,3 +51,16 @@ def test_forecast(client):
     assert len(rows) == 120
     assert {"Product", "SKU", "Date", "Demand", "Demand_Fitted"} <= \
         set(rows[0])
+
+
+@pytest.mark.gpu
+def test_classify_on_gpu_fused_path():
+    import torch
+    assert torch.cuda.is_available()
+    app = create_app(model_name="resnet18", num_classes=10, device="cuda:0")
+    with TestClient(app) as c:
+        img = np.random.default_rng(2).integers(
+            0, 256, (224, 224, 3), dtype=np.uint8)
+        r = c.post("/classify", json={"image": img.tolist(), "top_k": 4})
+        assert r.status_code == 200
+        assert len(r.json()["classes"]) == 4
