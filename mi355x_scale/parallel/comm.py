@@ -76,7 +76,15 @@ def assert_xgmi_mesh() -> None:
         if i != j and not ok
     ]
     if bad:
-        raise RuntimeError(f"peer access missing on GPU pairs: {bad}")
+        # RCCL still functions without peer access (host-staged, slow) —
+        # a degraded-topology box should produce a slow measured number,
+        # not kill the run. MI355X_REQUIRE_XGMI=1 restores the hard fail.
+        msg = f"peer access missing on GPU pairs: {bad}"
+        if os.environ.get("MI355X_REQUIRE_XGMI") == "1":
+            raise RuntimeError(msg)
+        import warnings
+        warnings.warn(f"{msg} — continuing with degraded RCCL transport",
+                      RuntimeWarning)
 
 
 def init_distributed(backend: Optional[str] = None,
