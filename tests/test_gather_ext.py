@@ -94,3 +94,31 @@ def test_scatter_oob_raises():
         _gather.scatter_f32(panel, np.array([5], dtype=np.int32),
                             np.array([0], dtype=np.int32),
                             np.array([1.0], dtype=np.float32))
+
+
+try:
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.lists(st.one_of(
+        st.text(min_size=0, max_size=12),
+        st.sampled_from(["dup1", "dup2", ""])), min_size=0, max_size=300))
+    def test_property_strings_match_pandas(vals):
+        col = pd.Series(vals, dtype=object)
+        codes, uniq = fast_factorize(col)
+        ref_codes, ref_uniq = pd.factorize(col, sort=True)
+        assert (codes == ref_codes).all()
+        assert list(uniq) == list(ref_uniq)
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.lists(st.integers(min_value=-2**62, max_value=2**62),
+                    min_size=0, max_size=300))
+    def test_property_ints_match_pandas(vals):
+        col = pd.Series(vals, dtype=np.int64)
+        codes, uniq = fast_factorize(col)
+        ref_codes, ref_uniq = pd.factorize(col, sort=True)
+        assert (codes == ref_codes).all()
+        assert (np.asarray(uniq) == np.asarray(ref_uniq)).all()
+except ImportError:  # pragma: no cover
+    pass
