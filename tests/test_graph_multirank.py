@@ -42,11 +42,21 @@ def _worker():
                     for p in model.parameters()])
     diag = {"after_ctor": _cross_rank_diff(p0)}
     for i in range(3):
-        gs.step(batch)
+        # manual step with forensics between the phases
+        for k, v in batch.items():
+            gs.static_batch[k].copy_(v, non_blocking=True)
+        gs.g_fwd_bwd.replay()
         torch.cuda.synchronize()
-        # the all-reduced flat grads are by construction the same sum on
-        # every rank; any divergence here is a collective bug
+        local_nan = int(gs.grad_buffers[0].isnan().sum().item())
+        both = [None, None]
+        dist.all_gather_object(both, local_nan)
+        diag[f"local_nan_{i}"] = tuple(both)
+        diag[f"loss_{i}"] = round(float(gs.static_loss.item()), 4)
+        gs._allreduce_grads()
         diag[f"grads_{i}"] = _cross_rank_diff(gs.grad_buffers[0])
+        if gs.g_opt is not None:
+            gs.g_opt.replay()
+        torch.cuda.synchronize()
         pp = torch.cat([p.detach().reshape(-1).cpu().float()
                         for p in model.parameters()])
         diag[f"params_{i}"] = _cross_rank_diff(pp)
@@ -60,4 +70,7 @@ def test_graph_step_two_ranks_one_gpu():
     # both ranks apply the identical all-reduced gradient every step, so
     # parameters stay in lockstep
     assert diag["after_ctor"] == 0.0, diag
-    assert all(v == 0.0 for v in diag.values()), diag
+    for i in range(3):
+        assert diag[f"local_nan_{i}"] == (0, 0), diag
+        assert diag[f"grads_{i}"] == 0.0, diag
+        assert diag[f"params_{i}"] == 0.0, diag
