@@ -80,9 +80,11 @@ class _FusedBNFn(torch.autograd.Function):
                 return g
             return None
 
+        import os
         wg, bg = _grad_view(weight), _grad_view(bias)
         fuse_acc = (torch.cuda.is_current_stream_capturing()
-                    and wg is not None and bg is not None)
+                    and wg is not None and bg is not None
+                    and os.environ.get("MI355X_BN_FUSED_ACC", "1") == "1")
         if fuse_acc:
             dweight, dbias = wg, bg
         else:
