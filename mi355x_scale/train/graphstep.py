@@ -81,6 +81,14 @@ class GraphedTrainStep:
         self.device = next(model.parameters()).device
         assert self.device.type == "cuda", "graph capture needs a GPU"
         self.autocast_dtype = autocast_dtype
+        # MIOpen find mode is REQUIRED under graph capture: immediate-mode
+        # can select conv solutions whose workspace preparation runs at
+        # find/capture time instead of in-stream, so the first replay is
+        # clean and every later replay reads the previous replay's
+        # workspace residue (observed as deterministic NaN conv weight
+        # grads from step 1 in the two-rank test). benchmark=True records
+        # the workspace-zeroing ops inside the captured graph.
+        torch.backends.cudnn.benchmark = True
 
         # Static input buffers (graph replays read these addresses).
         self.static_batch = {
