@@ -92,7 +92,10 @@ def init_distributed(backend: Optional[str] = None,
     rank, world, local = env_rank(), env_world_size(), env_local_rank()
     use_cuda = torch.cuda.is_available()
     if backend is None:
-        backend = "nccl" if use_cuda else "gloo"
+        # MI355X_BACKEND=gloo lets multi-rank paths be rehearsed with
+        # several ranks sharing one GPU (RCCL needs one rank per device)
+        backend = os.environ.get("MI355X_BACKEND") or (
+            "nccl" if use_cuda else "gloo")
     if use_cuda:
         torch.cuda.set_device(local % torch.cuda.device_count())
         device = torch.device("cuda", local % torch.cuda.device_count())
