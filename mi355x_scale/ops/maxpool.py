@@ -1,4 +1,5 @@
-"""Fused NHWC bf16 max-pool 3x3/stride-2/pad-1 (the ResNet stem pool).
+"""Fused NHWC bf16 max-pool 3x3/stride-2/pad-1 — the stem pool of the
+reference's ResNet family (``deep_learning/2...py:150``).
 
 Replaces torch's at::native NHWC max-pool pair, which cost 233 us fwd +
 517 us bwd per flagship step on MI355X and saves int64 indices (8 B per

@@ -1,5 +1,11 @@
 """hipGraph-captured training step.
 
+The MI355X replacement for the reference's Lightning DDP step
+(``deep_learning/2.distributed-data-loading-petastorm.py:390-397``):
+where Lightning lets NCCL bucket-hook ~400 eager kernel launches per
+step, here the whole step replays as two hipGraphs with one flat
+RCCL all-reduce between them.
+
 The profiled streaming bench showed the step is launch-bound: ~400 kernel
 launches per step from a GIL-contended Python thread leave the GPU ~45%
 idle (p50 15.5 ms, p95 80+ ms). The MI355X-idiomatic fix (north star:
