@@ -84,8 +84,10 @@ class Trainer:
         for key, vals in self._sync_accum.items():
             mean = sum(v for v, _ in vals) / len(vals)
             if any(s for _, s in vals) and dist.is_initialized():
+                # tensor device follows the BACKEND (gloo reduces on host
+                # even when compute is on GPU)
                 t = torch.tensor([mean], device=self.ctx.device
-                                 if self.ctx.device.type == "cuda" else "cpu")
+                                 if dist.get_backend() == "nccl" else "cpu")
                 dist.all_reduce(t, op=dist.ReduceOp.SUM)
                 mean = (t / dist.get_world_size()).item()
             out[key] = mean
