@@ -107,6 +107,9 @@ class Run:
                     f"final_status: {status}\n")
 
     def __enter__(self):
+        # MLflow layout: nested runs carry the parent run id as a tag
+        if _active_run_stack:
+            self.set_tag("mlflow.parentRunId", _active_run_stack[-1].run_id)
         _active_run_stack.append(self)
         return self
 

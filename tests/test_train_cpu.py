@@ -63,3 +63,15 @@ def test_trainer_resume(image_parquet, tmp_path):
                  resume_from=str(tmp_path / "last.ckpt"))
     t2.fit(model2, dm2)
     assert model2.global_step > 2  # continued past the restored step
+
+
+def test_nested_runs_carry_parent_tag(tmp_path):
+    from mi355x_scale import track
+    track.set_tracking_root(str(tmp_path / "mlruns"))
+    track.set_experiment("nested-test")
+    with track.start_run("outer") as outer:
+        with track.start_run("inner", nested=True) as inner:
+            inner.log_metric("loss", 1.0)
+        tag = (tmp_path / "mlruns").rglob("mlflow.parentRunId")
+        [tagfile] = list(tag)
+        assert tagfile.read_text() == outer.run_id
