@@ -92,3 +92,31 @@ def test_trainer_hipgraph_mode(tmp_path):
                       default_root_dir=str(tmp_path / "ckpt"))
     trainer.fit(model, dm)
     assert os.path.exists(tmp_path / "ckpt" / "last.ckpt")
+
+
+@pytest.mark.gpu
+def test_trainer_graph_mode_resume(tmp_path):
+    """Graph-mode checkpoints (FlatAdam flat-buffer format) must resume:
+    a second Trainer picks up epoch/step and keeps training."""
+    import os
+    from mi355x_scale.data.generator import write_image_parquet
+    from mi355x_scale.train import ImageStreamDataModule, Trainer
+    d = str(tmp_path / "imgs")
+    write_image_parquet(d, num_rows=64, image_hw=(64, 64), num_classes=10,
+                        rows_per_group=16, rows_per_file=32)
+    root = str(tmp_path / "ckpt")
+    model = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
+    dm = ImageStreamDataModule(d, batch_size=16, workers_count=2,
+                               image_hw=(64, 64))
+    Trainer(max_epochs=1, limit_train_batches=3, limit_val_batches=1,
+            default_root_dir=root).fit(model, dm)
+    assert os.path.exists(os.path.join(root, "last.ckpt"))
+
+    model2 = ImageClassifier("resnet18", num_classes=10, lr=1e-3)
+    dm2 = ImageStreamDataModule(d, batch_size=16, workers_count=2,
+                                image_hw=(64, 64))
+    t2 = Trainer(max_epochs=2, limit_train_batches=3, limit_val_batches=1,
+                 default_root_dir=root,
+                 resume_from=os.path.join(root, "last.ckpt"))
+    out = t2.fit(model2, dm2)
+    assert out.global_step >= 6  # continued past the first run's steps
