@@ -24,3 +24,10 @@ def test_resnet18_small_input():
     m = resnet18(num_classes=10)
     y = m(torch.randn(2, 3, 64, 64))
     assert y.shape == (2, 10)
+
+
+def test_resnet34_params():
+    from mi355x_scale.models import resnet34
+    m = resnet34(num_classes=1000)
+    n_params = sum(p.numel() for p in m.parameters())
+    assert n_params == 21_797_672  # canonical ResNet-34/1000 size
