@@ -163,21 +163,24 @@ def generate_bom(skus: List[str], levels: int = 3,
         root = f"M_{sku}"
         mapper.append({"sku": sku, "final_mat_number": root})
         frontier = [root]
+        sku_edges = []
         for _ in range(levels):
             nxt = []
             for node in frontier:
                 k = int(rng.integers(children[0], children[1] + 1))
                 for child in _rand_ids(rng, k):
                     qty = int(rng.integers(1, 5))
-                    edges.append({"material_in": child, "material_out": node,
-                                  "qty": qty})
+                    sku_edges.append({"material_in": child,
+                                      "material_out": node, "qty": qty})
                     nxt.append(child)
             frontier = nxt
+        # Validate THIS SKU's subgraph only (O(edges-per-SKU), not the
+        # cumulative edge list of every SKU generated so far).
         g = nx.DiGraph()
         g.add_edges_from((e["material_in"], e["material_out"])
-                         for e in edges if e["material_out"].startswith(f"M_{sku}")
-                         or True)
+                         for e in sku_edges)
         assert nx.is_directed_acyclic_graph(g)
+        edges.extend(sku_edges)
     return pd.DataFrame(edges), pd.DataFrame(mapper)
 
 

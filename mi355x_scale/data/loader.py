@@ -57,7 +57,6 @@ class DataLoader:
         self.reader = reader
         self.batch_size = int(batch_size)
         self.drop_last = drop_last
-        self._carry: Optional[Dict[str, np.ndarray]] = None
 
     def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
         carry: Optional[Dict[str, np.ndarray]] = None
@@ -245,7 +244,15 @@ class DeviceLoader:
                         return
                     continue
                 ev, dev = item
-                torch.cuda.current_stream(self.device).wait_event(ev)
+                cur = torch.cuda.current_stream(self.device)
+                cur.wait_event(ev)
+                # The device tensors were allocated on a stager's copy
+                # stream; tell the caching allocator the consumer stream
+                # uses them, or freeing a batch mid-flight lets the block
+                # be recycled for the next H2D while the compute stream is
+                # still reading it.
+                for t in dev.values():
+                    t.record_stream(cur)
                 yield dev
         finally:
             stop.set()

@@ -16,6 +16,7 @@ so MLflow UIs/tools pointed at the directory can read the runs.
 """
 from __future__ import annotations
 
+import hashlib
 import os
 import time
 import uuid
@@ -46,9 +47,15 @@ def set_tracking_root(root: str) -> None:
 
 
 def set_experiment(name: str, experiment_id: Optional[str] = None) -> None:
-    """MLflow-compat: select/create the experiment runs land in."""
+    """MLflow-compat: select/create the experiment runs land in.
+
+    The id is derived deterministically from the name (md5, not Python's
+    per-process salted ``hash``) so every process/run maps the same
+    experiment name to the same ``mlruns/<id>/`` directory.
+    """
     _active_experiment["name"] = name
-    _active_experiment["id"] = experiment_id or str(abs(hash(name)) % 10**9)
+    _active_experiment["id"] = experiment_id or str(
+        int(hashlib.md5(name.encode()).hexdigest()[:12], 16) % 10**9)
     _exp_dir()
 
 

@@ -69,6 +69,12 @@ class CheckpointManager:
         state = torch.load(path, map_location=map_location,
                            weights_only=False)
         model.load_state_dict(state["model"])
-        if optimizer is not None and state.get("optimizer"):
-            optimizer.load_state_dict(state["optimizer"])
+        if optimizer is not None:
+            if state.get("optimizer"):
+                optimizer.load_state_dict(state["optimizer"])
+            elif hasattr(optimizer, "sync_master_from_params"):
+                # weights-only checkpoint: re-seed the optimizer's fp32
+                # master from the freshly loaded (bf16) param views, or
+                # the first step() reverts the model to its init values
+                optimizer.sync_master_from_params()
         return state

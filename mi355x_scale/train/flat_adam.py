@@ -81,6 +81,10 @@ class FlatAdam:
             self.flat_gb16 = torch.zeros(0, dtype=torch.bfloat16,
                                          device=device)
 
+        # (param, grad-buffer key, offset into that buffer, numel) — the
+        # comm-overlap path (train/graphstep.py) uses this to find the
+        # flat-buffer split offsets for a given backward bucket boundary
+        self.param_layout: List[Tuple[torch.nn.Parameter, str, int, int]] = []
         off = 0
         for p in bf16_set:
             n = p.numel()
@@ -95,6 +99,7 @@ class FlatAdam:
             p.data = pview
             p.grad = self.flat_gb16[off:off + n].as_strided(
                 p.shape, p.stride())
+            self.param_layout.append((p, "bf16", off, n))
             off += n
         foff = 0
         for p in fp32_set:
@@ -105,6 +110,7 @@ class FlatAdam:
             p.data = pview
             p.grad = self.flat_grads[foff:foff + n].as_strided(
                 p.shape, p.stride())
+            self.param_layout.append((p, "f32", foff, n))
             off += n
             foff += n
 
@@ -168,6 +174,49 @@ class FlatAdam:
         self.flat_master.addcdiv_(self.exp_avg, denom, value=-self.lr / bc1)
         if self.n_bf16:
             self.flat_pb16.copy_(self.flat_master[:self.n_bf16])
+
+    def split_offsets(self, late_params) -> dict:
+        """Per-grad-buffer split offset for a two-bucket all-reduce.
+
+        ``late_params``: the params whose grads are ready after the FIRST
+        backward stage (the layers closest to the loss). Returns
+        ``{"bf16": off, "f32": off}`` such that each buffer's
+        ``[off:]`` tail holds exactly the late params' grads. Raises if
+        the late set is not a contiguous tail of each buffer (the flat
+        layout follows registration order, so any suffix of the model's
+        parameter list is).
+        """
+        late = {id(p) for p in late_params}
+        splits = {}
+        for key in ("bf16", "f32"):
+            entries = [(o, n, id(p) in late)
+                       for p, k, o, n in self.param_layout if k == key]
+            if not entries:
+                continue
+            tail = [o for o, n, is_late in entries if is_late]
+            split = min(tail) if tail else (
+                max(o + n for o, n, _ in entries))
+            for o, n, is_late in entries:
+                if is_late != (o >= split):
+                    raise ValueError(
+                        "late params are not a contiguous tail of the "
+                        f"{key} grad buffer — comm overlap needs a "
+                        "boundary that splits the registration order")
+            splits[key] = split
+        return splits
+
+    @torch.no_grad()
+    def sync_master_from_params(self) -> None:
+        """Copy the current parameter values back into the fp32 master.
+
+        Needed after a weights-only resume: ``model.load_state_dict``
+        writes the bf16 working views, and without this the master (which
+        the next ``step()`` treats as canonical) silently keeps its
+        construction-time values. fp32 params are views INTO the master,
+        so only the bf16 segment needs the copy-back.
+        """
+        if self.n_bf16:
+            self.flat_master[:self.n_bf16].copy_(self.flat_pb16)
 
     def zero_grad(self, set_to_none: bool = False) -> None:
         # grads are views of the flat buffers; zeroing the buffers is the

@@ -16,10 +16,19 @@ hipGraph and replay it with a single launch per step.
 Data parallism is done manually around the graph (DDP's bucketed hooks
 can't fire inside a replay): gradients live in ONE flat buffer
 (``p.grad`` are views), the all-reduce is a single RCCL call on that
-buffer between the fwd/bwd graph and the optimizer graph. For ResNet-18
-(~45 MB fp32 grads) one flat all-reduce over xGMI beats DDP's bucket
-pipeline — there is almost no backward left to overlap with after graph
-capture.
+buffer between the fwd/bwd graph and the optimizer graph.
+
+Comm/compute overlap (round 2): with a boundary module declared
+(``model.comm_overlap_boundary``), the backward is split in two captured
+graphs at that activation — the classic pipeline cut: the boundary
+module's output is detached into a leaf, so ``loss.backward()`` stops
+there after producing the LATE layers' grads (the tail of the flat
+buffers, ~94% of ResNet-18's bytes), and a second graph continues
+``a.backward(a2.grad)`` through the early layers. Between the two
+replays the tail all-reduce runs on a dedicated comm stream, overlapped
+with the early-backward replay — the overlap DDP's bucket hooks give the
+reference for free (``deep_learning/2.distributed-data-loading-
+petastorm.py:390-397``), bought back under graph capture.
 """
 from __future__ import annotations
 
@@ -131,13 +140,31 @@ class GraphedTrainStep:
         self.flat_grads = self.grad_buffers[0]  # back-compat alias
         self._ar_states = [dict() for _ in self.grad_buffers]
 
-        def _fwd_bwd():
+        # Grad buffers by layout key (FlatAdam only) — the comm-overlap
+        # path needs named buffers + per-param offsets to split them.
+        self._buf_by_key = {}
+        if hasattr(optimizer, "param_layout"):
+            if getattr(optimizer, "n_bf16", 0):
+                self._buf_by_key["bf16"] = optimizer.flat_gb16
+            if optimizer.flat_grads.numel():
+                self._buf_by_key["f32"] = optimizer.flat_grads
+        self._overlap = self._setup_overlap(model, optimizer, world_size)
+        self.g_bwd2 = None
+        holder = self._overlap["holder"] if self._overlap else None
+
+        def _fwd():
             for b in self.grad_buffers:
                 b.zero_()
             with torch.autocast(device_type="cuda", dtype=autocast_dtype,
                                 enabled=autocast_dtype is not None):
-                loss = model.training_step(self.static_batch, 0)
+                return model.training_step(self.static_batch, 0)
+
+        def _fwd_bwd():
+            loss = _fwd()
             loss.backward()
+            if self._overlap is not None:
+                # continue through the early layers from the cut point
+                holder["a"].backward(holder["a2"].grad)
             return loss
 
         # Warmup on a side stream (MIOpen find, allocator steady state)
@@ -156,14 +183,30 @@ class GraphedTrainStep:
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
 
-        # Capture forward+backward (one graph). thread_local capture
-        # mode: the streaming loader's stager threads keep issuing H2D
-        # copies on their own streams during capture — global mode would
-        # abort the process on their first HIP call.
+        # Capture forward+backward. thread_local capture mode: the
+        # streaming loader's stager threads keep issuing H2D copies on
+        # their own streams during capture — global mode would abort the
+        # process on their first HIP call.
         self.g_fwd_bwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwd_bwd,
                               capture_error_mode="thread_local"):
-            self.static_loss = _fwd_bwd()
+            self.static_loss = _fwd()
+            # with the cut installed this stops at the boundary leaf,
+            # producing only the LATE layers' grads (buffer tails)
+            self.static_loss.backward()
+        if self._overlap is not None:
+            # capture-time tensors: the hook ran inside the capture above,
+            # so a/a2 (and a2.grad, written fresh by graph A's recorded
+            # kernels — .grad was None at capture, no accumulation) live
+            # in graph A's private pool; graph B must share that pool.
+            a, a2 = holder["a"], holder["a2"]
+            self.g_bwd2 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_bwd2, pool=self.g_fwd_bwd.pool(),
+                                  capture_error_mode="thread_local"):
+                a.backward(a2.grad)
+            # replays never run python hooks; remove so eager validation
+            # forwards are not rerouted through the detached leaf
+            self._overlap["hook_handle"].remove()
 
         # ...and the optimizer step (second graph; the eager RCCL
         # all-reduce slots between the two replays when world_size > 1).
@@ -183,6 +226,64 @@ class GraphedTrainStep:
                 optimizer.step()
             self.g_opt = g
 
+    def _setup_overlap(self, model, optimizer, world_size):
+        """Install the boundary cut for split-backward comm overlap.
+
+        Enabled when the module declares ``comm_overlap_boundary`` (a
+        submodule path whose OUTPUT activation is the cut), the optimizer
+        exposes a flat layout, and either world_size > 1 or
+        ``MI355X_GRAPH_OVERLAP=1`` forces it (the 1-GPU parity test).
+        ``MI355X_GRAPH_OVERLAP=0`` or the hand p2p all-reduce path
+        disable it (p2p stays monolithic).
+        """
+        import os
+        flag = os.environ.get("MI355X_GRAPH_OVERLAP", "auto")
+        if flag == "0":
+            return None
+        if world_size <= 1 and flag != "1":
+            return None
+        if os.environ.get("MI355X_P2P_ALLREDUCE") == "1":
+            return None
+        boundary = getattr(model, "comm_overlap_boundary", None)
+        if boundary is None or not self._buf_by_key:
+            return None
+        try:
+            mod = model.get_submodule(boundary)
+        except AttributeError:
+            return None
+        named = list(model.named_parameters())
+        prefix = boundary + "."
+        idxs = [i for i, (n, _) in enumerate(named)
+                if n.startswith(prefix)]
+        if not idxs or idxs[-1] + 1 >= len(named):
+            return None
+        late_params = [p for _, p in named[idxs[-1] + 1:]
+                       if p.requires_grad]
+        try:
+            splits = optimizer.split_offsets(late_params)
+        except ValueError:
+            return None
+
+        holder: Dict[str, torch.Tensor] = {}
+
+        def _cut(module, inputs, out):
+            holder["a"] = out
+            a2 = out.detach().requires_grad_(True)
+            holder["a2"] = a2
+            return a2
+
+        return {
+            "holder": holder,
+            "hook_handle": mod.register_forward_hook(_cut),
+            "splits": splits,
+            "stream": torch.cuda.Stream(device=self.device),
+            "ev_a": torch.cuda.Event(),
+            "ev_b": torch.cuda.Event(),
+            "ev_done": torch.cuda.Event(),
+            "tail_states": {k: {} for k in self._buf_by_key},
+            "head_states": {k: {} for k in self._buf_by_key},
+        }
+
     def _allreduce_grads(self) -> None:
         for b, st in zip(self.grad_buffers, self._ar_states):
             allreduce_flat(b, self.world_size, st)
@@ -191,8 +292,36 @@ class GraphedTrainStep:
         for k, v in batch.items():
             self.static_batch[k].copy_(v, non_blocking=True)
         self.g_fwd_bwd.replay()
-        if self.world_size > 1:
-            self._allreduce_grads()
+        if self.g_bwd2 is not None and self.world_size > 1:
+            # tail (late-layer) all-reduce on the comm stream, overlapped
+            # with the early-backward replay on the compute stream
+            ov = self._overlap
+            cur = torch.cuda.current_stream(self.device)
+            comm = ov["stream"]
+            ov["ev_a"].record(cur)
+            comm.wait_event(ov["ev_a"])
+            with torch.cuda.stream(comm):
+                for key, buf in self._buf_by_key.items():
+                    sp = ov["splits"].get(key, 0)
+                    if sp < buf.numel():
+                        allreduce_flat(buf[sp:], self.world_size,
+                                       ov["tail_states"][key])
+            self.g_bwd2.replay()
+            ov["ev_b"].record(cur)
+            comm.wait_event(ov["ev_b"])
+            with torch.cuda.stream(comm):
+                for key, buf in self._buf_by_key.items():
+                    sp = ov["splits"].get(key, 0)
+                    if sp > 0:
+                        allreduce_flat(buf[:sp], self.world_size,
+                                       ov["head_states"][key])
+            ov["ev_done"].record(comm)
+            cur.wait_event(ov["ev_done"])
+        else:
+            if self.g_bwd2 is not None:
+                self.g_bwd2.replay()  # split capture, single rank
+            if self.world_size > 1:
+                self._allreduce_grads()
         if self.g_opt is not None:
             self.g_opt.replay()
         else:

@@ -57,6 +57,11 @@ class ImageClassifier(ScaleModule):
     135-208``): backbone + Adam + cross-entropy train step + val step
     logging loss and multiclass accuracy."""
 
+    # split-backward comm overlap cut (train/graphstep.py): backward of
+    # layer3/layer4/fc (~94% of grad bytes) completes first, so its
+    # all-reduce overlaps the early-layer backward replay
+    comm_overlap_boundary = "model.layer2"
+
     def __init__(self, model_name: str = "resnet18", num_classes: int = 1000,
                  lr: float = 1e-5, channels_last: bool = True):
         super().__init__()

@@ -125,11 +125,32 @@ class Trainer:
 
         if graph_on:
             # fused flat Adam: one HIP kernel per step; its flat grad
-            # buffer is the graphed step's single all-reduce target
+            # buffer is the graphed step's single all-reduce target.
+            # Hyperparameters come from the module's own
+            # configure_optimizers() when it builds an Adam — graph mode
+            # swaps the implementation, not the optimizer settings.
             from .flat_adam import FlatAdam
+            hp = {"lr": getattr(model, "lr", 1e-3),
+                  "betas": (0.9, 0.999), "eps": 1e-8, "weight_decay": 0.0}
+            try:
+                configured = model.configure_optimizers()
+            except Exception:
+                configured = None
+            if configured is not None and getattr(
+                    configured, "param_groups", None):
+                g0 = configured.param_groups[0]
+                for k in hp:
+                    if k in g0:
+                        hp[k] = g0[k]
+                if not isinstance(configured, torch.optim.Adam):
+                    import warnings
+                    warnings.warn(
+                        "graph mode replaces the configured "
+                        f"{type(configured).__name__} with FlatAdam "
+                        "(Adam math); pass use_hipgraph='off' to keep it")
+                del configured
             optimizer = FlatAdam(model.parameters(),
-                                 lr=getattr(model, "lr", 1e-3),
-                                 bf16_params=device.type == "cuda")
+                                 bf16_params=device.type == "cuda", **hp)
         else:
             optimizer = model.configure_optimizers()
         start_epoch = 0
@@ -214,11 +235,43 @@ class Trainer:
         return model
 
     @torch.no_grad()
+    def _broadcast_buffers(self, model) -> None:
+        """Rank-0 → all broadcast of module buffers (BN running stats).
+
+        DDP broadcasts buffers every forward; the graphed step does not
+        touch them cross-rank, so running_mean/var drift per rank (train
+        forwards use batch stats — training is unaffected) and eval
+        metrics would disagree. One coalesced broadcast before each
+        validation pass restores DDP's eval semantics.
+        """
+        if not (dist.is_initialized() and dist.get_world_size() > 1):
+            return
+        bufs = [b for b in model.buffers() if b.numel()]
+        if not bufs:
+            return
+        # coalesce per dtype to avoid one collective per BN layer
+        by_dtype: dict = {}
+        for b in bufs:
+            by_dtype.setdefault(b.dtype, []).append(b)
+        on_host = dist.get_backend() == "gloo"
+        for dt, group in by_dtype.items():
+            flat = torch.cat([b.detach().reshape(-1) for b in group])
+            if on_host:
+                flat = flat.cpu()
+            dist.broadcast(flat, src=0)
+            flat = flat.to(group[0].device)
+            off = 0
+            for b in group:
+                b.copy_(flat[off:off + b.numel()].view_as(b))
+                off += b.numel()
+
+    @torch.no_grad()
     def _run_validation(self, wrapped, model, datamodule, device, amp_dtype,
                         autocast_on, step) -> dict:
         loader = datamodule.val_dataloader()
         if loader is None:
             return {}
+        self._broadcast_buffers(model)
         (wrapped or model).eval()
         limit = self.limit_val_batches or float("inf")
         n = 0

@@ -249,4 +249,19 @@ def fmin(fn: Callable, space, algo=None, max_evals: int = 10,
                     done_n += 1
 
     best = trials.best_trial
-    return {k: v[0] for k, v in best["misc"]["vals"].items()}
+    vals = {k: v[0] for k, v in best["misc"]["vals"].items()}
+    # hyperopt returns option INDICES for hp.choice parameters (so
+    # reference-style `options[best[key]]` / `space_eval(space, best)`
+    # works); internal trial storage keeps raw values — the one place the
+    # index convention surfaces is this return dict.
+    from .space import Choice, IntCast, flatten_space
+    flat = flatten_space(space)
+    out = {}
+    for k, v in vals.items():
+        expr = flat.get(k)
+        base = expr.inner if isinstance(expr, IntCast) else expr
+        if isinstance(base, Choice):
+            out[k] = int(base.to_internal(v))
+        else:
+            out[k] = v
+    return out

@@ -261,6 +261,23 @@ def flatten_space(space) -> Dict[str, Expr]:
     return out
 
 
+def space_eval(space, best: Dict[str, Any]):
+    """hyperopt-compat: substitute ``fmin``'s best dict into the space.
+
+    ``fmin`` returns option *indices* for ``hp.choice`` parameters (as
+    hyperopt does); this maps them back to the option values and rebuilds
+    the user's space structure.
+    """
+    values: Dict[str, Any] = {}
+    for lbl, expr in flatten_space(space).items():
+        v = best[lbl]
+        base = expr.inner if isinstance(expr, IntCast) else expr
+        if isinstance(base, Choice):
+            v = base.options[int(v)]
+        values[lbl] = v
+    return bind_params(space, values)
+
+
 def bind_params(space, values: Dict[str, Any]):
     """Rebuild the user's space structure with sampled values in place."""
     if isinstance(space, Expr):
