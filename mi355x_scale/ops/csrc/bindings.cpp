@@ -55,7 +55,10 @@ extern "C" void launch_groupfit_final(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
     const float* pj0, const float* pj1, const float* pj2,
     const int* best_order, float* fitted, float* params,
-    unsigned char* statusv, int T, long long G, int KX, hipStream_t stream);
+    unsigned char* statusv,
+    const float* b0, const float* b1, const float* b2,
+    const float* w0, const float* w1, const float* w2,
+    int T, long long G, int KX, hipStream_t stream);
 
 static void _check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
@@ -99,7 +102,9 @@ void groupfit_final(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
                     torch::Tensor xc2, torch::Tensor pj0, torch::Tensor pj1,
                     torch::Tensor pj2, torch::Tensor best_order,
                     torch::Tensor fitted, torch::Tensor params,
-                    torch::Tensor status) {
+                    torch::Tensor status,
+                    torch::Tensor b0, torch::Tensor b1, torch::Tensor b2,
+                    torch::Tensor w0, torch::Tensor w1, torch::Tensor w2) {
   _check_f32(yT, "yT"); _check_f32(fitted, "fitted");
   _check_f32(params, "params");
   TORCH_CHECK(best_order.scalar_type() == torch::kInt32 &&
@@ -108,12 +113,17 @@ void groupfit_final(torch::Tensor yT, torch::Tensor xc0, torch::Tensor xc1,
   long long G = yT.size(1);
   TORCH_CHECK(fitted.size(0) == T && fitted.size(1) == G);
   auto stream = at::cuda::getCurrentHIPStream();
+  auto fp = [](torch::Tensor& t) -> const float* {
+    return t.numel() ? t.data_ptr<float>() : nullptr;
+  };
   launch_groupfit_final(
       yT.data_ptr<float>(), xc0.data_ptr<float>(), xc1.data_ptr<float>(),
       xc2.data_ptr<float>(), pj0.data_ptr<float>(), pj1.data_ptr<float>(),
       pj2.data_ptr<float>(), best_order.data_ptr<int>(),
       fitted.data_ptr<float>(), params.data_ptr<float>(),
-      status.data_ptr<uint8_t>(), T, G, (int)xc0.size(1), stream.stream());
+      status.data_ptr<uint8_t>(),
+      fp(b0), fp(b1), fp(b2), fp(w0), fp(w1), fp(w2),
+      T, G, (int)xc0.size(1), stream.stream());
 }
 
 // Wc [n][G], wm [G] from the time-major panel (differenced d times,

@@ -5,7 +5,7 @@
 // with a fixed-schedule estimator (see forecast/batched.py — the numpy
 // oracle these kernels are tested against):
 //   stage 1: OLS regression on shared exog design (host-precomputed
-//            pseudo-inverse -> per-group matvec)
+//            pseudo-inverse / MFMA design GEMM -> per-group beta)
 //   stage 2: Yule-Walker + Levinson-Durbin long-AR -> innovations
 //   stage 3: lag-matrix normal equations (<=8x8), ridge + Cholesky,
 //            stationarity shrinkage; one refinement pass (stage 4)
@@ -13,19 +13,24 @@
 //
 // MI355X-first decomposition: ONE GROUP PER LANE, 64 groups per wave,
 // one wave per workgroup. Global series are TIME-MAJOR [T][G] so every
-// per-timestep access is a fully coalesced 64-lane load. Per-group
-// series (u, eps) live in LDS slabs [n][64] (bank = (t*64+lane)%64 —
-// all 64 lanes hit distinct banks every access); per-group small
-// matrices/vectors (A, r, coefs) live in a per-lane LDS state block
-// because runtime-indexed register arrays would spill to scratch
-// (guide §5.4 rule 20). Loop trip counts are uniform across the wave in
-// the eval kernel (one candidate order at a time), so the hot loops have
-// no divergence; the final-fit kernel predicates on per-lane (p,d,q).
+// per-timestep access is a fully coalesced 64-lane load.
+//
+// STREAMING (round 2): every stage has <= 8-lag lookback, so the u/eps
+// series never materialize. Each pass re-derives u_t on the fly from
+// the y stream (1 coalesced load + KX fma) and carries 8-deep per-lane
+// REGISTER rings (constant-indexed under full unrolls — runtime-indexed
+// register arrays would spill, guide §5.4). The round-1 version staged
+// u/eps in LDS slabs (2 x S x 64 lanes fp32 ≈ 60 KB/block) and ran ONE
+// wave per CU; dropping the slabs leaves only the per-lane state block
+// (small matrices/vectors, 120 floats x 64 lanes = 30 KB) and lifts
+// occupancy to ~5 waves/CU. The extra y re-reads are noise: ~4 passes
+// x n x 4 B x G per candidate ≈ 190 MB at 8 TB/s HBM3E.
 //
 // Design note on MFMA: the per-group normal equations are at most 8x8
 // built from ~120-step series — batching 64 groups per wave makes the
-// whole fit bandwidth/latency-bound (µs-scale per candidate for 100k
-// groups), so matrix cores have nothing to accelerate in this op.
+// whole fit latency-bound, so matrix cores only help stage 1 (the
+// [KX,n]x[n,G] design GEMM, ops/csrc/mfma_project.hip), which both the
+// eval and final kernels take precomputed.
 
 #include <hip/hip_runtime.h>
 #include <math.h>
@@ -106,144 +111,250 @@ __device__ void levinson(float* st, int lane, int M) {
   }
 }
 
-// Core fit for one (group-lane, order): fills U (regression residuals)
-// and E (innovations) slabs and the ST_BETA/ST_PHI/ST_THETA state.
-// Returns the mean of the differenced series; *ok_out false on
-// non-finite state.
-__device__ __forceinline__ float fit_lane(
+// ---------------------------------------------------------------------
+// Streaming pass machinery. Each pass walks t = 0..S-1, re-deriving
+//   w_t  = diff^d(y)_t            (y0/y1 register ring)
+//   u_td = (w_t - wm) - x_td . beta
+// and carries ur[j] = u_{td-1-j}, er[j] = e_{td-1-j} register rings
+// (zero-filled: pre-sample lags are zero, matching the oracle).
+// ---------------------------------------------------------------------
+
+#define RING_SHIFT(r, v)                        \
+  _Pragma("unroll") for (int _j = 7; _j > 0; --_j) r[_j] = r[_j - 1]; \
+  r[0] = (v)
+
+// One streaming u-producer step. Returns u_td (only valid when t >= d).
+__device__ __forceinline__ float u_step(
     const float* __restrict__ yT, const float* __restrict__ xc,
-    const float* __restrict__ PJ, float* U, float* E, float* st,
+    const float* st, int lane, long long g, long long G, int KX,
+    int d, int t, float wm, float* y0, float* y1) {
+  const float yv = yT[(long long)t * G + g];
+  float w;
+  if (d == 0) w = yv;
+  else if (d == 1) w = yv - *y0;
+  else w = yv - 2.0f * (*y0) + (*y1);
+  *y1 = *y0; *y0 = yv;
+  if (t < d) return 0.0f;
+  const int td = t - d;
+  float reg = 0.0f;
+  for (int k = 0; k < KX; ++k)
+    reg += xc[(long long)td * KX + k] * st[SIDX(ST_BETA + k)];
+  return (w - wm) - reg;
+}
+
+// Core fit for one (group-lane, order): leaves ST_BETA/ST_PHI/ST_THETA
+// state and the end-of-series rings ur/er (u_{n-1-j}, e_{n-1-j}).
+// *ok_out false on non-finite state.
+__device__ float fit_lane(
+    const float* __restrict__ yT, const float* __restrict__ xc,
+    const float* __restrict__ PJ, float* st,
     int lane, long long g, long long G, int S, int KX, int p, int d, int q,
-    bool* ok_out,
+    bool* ok_out, float* ur, float* er,
     const float* __restrict__ beta_pre,  // [KX][G] from the MFMA
     const float* __restrict__ wm_pre) {  // [G]     projection, or null
   const int n = S - d;
-  // ---- stage 0: difference + mean (stream y, write w into U)
-  float y0 = 0.f, y1 = 0.f;
-  float wsum = 0.0f;
-  for (int t = 0; t < S; ++t) {
-    float yv = yT[(long long)t * G + g];
-    float w;
-    if (d == 0) w = yv;
-    else if (d == 1) w = yv - y0;
-    else w = yv - 2.0f * y0 + y1;
-    y1 = y0; y0 = yv;
-    if (t >= d) {
-      U[(t - d) * WAVE + lane] = w;
-      wsum += w;
+  float y0, y1;
+
+  // ---- stage 0: mean of the differenced series (pass only if needed)
+  float wm;
+  if (wm_pre) {
+    wm = wm_pre[g];
+  } else {
+    float wsum = 0.0f;
+    y0 = y1 = 0.0f;
+    for (int t = 0; t < S; ++t) {
+      const float yv = yT[(long long)t * G + g];
+      float w;
+      if (d == 0) w = yv;
+      else if (d == 1) w = yv - y0;
+      else w = yv - 2.0f * y0 + y1;
+      y1 = y0; y0 = yv;
+      if (t >= d) wsum += w;
     }
+    wm = wsum / (float)n;
   }
-  const float wm = wm_pre ? wm_pre[g] : (wsum / (float)n);
-  // ---- stage 1: beta = P @ (w - wm); u = wc - Xc beta. beta comes from
-  // the MFMA design-matrix GEMM (ops/csrc/mfma_project.hip) when
-  // precomputed, else from the per-lane matvec.
+
+  // ---- stage 1: beta = P @ (w - wm) (pass only if not precomputed)
 #pragma unroll
   for (int k = 0; k < MAXK; ++k) BETA(k) = 0.0f;
   if (beta_pre) {
     for (int k = 0; k < KX; ++k) BETA(k) = beta_pre[(long long)k * G + g];
   } else {
-    for (int k = 0; k < KX; ++k) {
-      float acc = 0.0f;
-      const float* Pk = PJ + (long long)k * n;
-      for (int t = 0; t < n; ++t)
-        acc += Pk[t] * (U[t * WAVE + lane] - wm);
-      BETA(k) = acc;
+    float bacc[MAXK];
+#pragma unroll
+    for (int k = 0; k < MAXK; ++k) bacc[k] = 0.0f;
+    y0 = y1 = 0.0f;
+    for (int t = 0; t < S; ++t) {
+      const float yv = yT[(long long)t * G + g];
+      float w;
+      if (d == 0) w = yv;
+      else if (d == 1) w = yv - y0;
+      else w = yv - 2.0f * y0 + y1;
+      y1 = y0; y0 = yv;
+      if (t < d) continue;
+      const int td = t - d;
+      const float wcv = w - wm;
+#pragma unroll
+      for (int k = 0; k < MAXK; ++k)
+        if (k < KX) bacc[k] += PJ[(long long)k * n + td] * wcv;
     }
+#pragma unroll
+    for (int k = 0; k < MAXK; ++k)
+      if (k < KX) BETA(k) = bacc[k];
   }
-  for (int t = 0; t < n; ++t) {
-    float reg = 0.0f;
-    for (int k = 0; k < KX; ++k) reg += xc[(long long)t * KX + k] * BETA(k);
-    U[t * WAVE + lane] = (U[t * WAVE + lane] - wm) - reg;
-  }
-  // ---- stage 2: long-AR innovations
-  if (q > 0) {
-    int M = max(p, q) + 3;
-    if (M > MAXM) M = MAXM;
-    if (M > n / 4) M = max(1, n / 4);
-    for (int k = 0; k <= M; ++k) {
-      float acc = 0.0f;
-      for (int t = k; t < n; ++t)
-        acc += U[t * WAVE + lane] * U[(t - k) * WAVE + lane];
-      st[SIDX(ST_R + k)] = acc / (float)n;
-    }
-    levinson(st, lane, M);
-    for (int t = 0; t < n; ++t) {
-      float acc = U[t * WAVE + lane];
-      for (int i = 1; i <= M; ++i)
-        if (t - i >= 0)
-          acc -= st[SIDX(ST_AL + (i - 1))] * U[(t - i) * WAVE + lane];
-      E[t * WAVE + lane] = acc;
-    }
-  } else {
-    for (int t = 0; t < n; ++t) E[t * WAVE + lane] = U[t * WAVE + lane];
-  }
-  // ---- stages 3+4: lag OLS, refine once
+
 #pragma unroll
   for (int i = 0; i < 4; ++i) { PHI(i) = 0.0f; THETA(i) = 0.0f; }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) { ur[j] = 0.0f; er[j] = 0.0f; }
+
   const int K = p + q;
-  if (K > 0) {
-    const int m = max(max(p, q), 1);
-    for (int pass = 0; pass < 2; ++pass) {
-      for (int i = 0; i < 64; ++i) st[SIDX(ST_A + i)] = 0.0f;
-      for (int i = 0; i < 8; ++i) st[SIDX(ST_C + i)] = 0.0f;
-      for (int t = m; t < n; ++t) {
-        for (int i = 0; i < p; ++i)
-          st[SIDX(ST_Z + i)] = U[(t - 1 - i) * WAVE + lane];
-        for (int j = 0; j < q; ++j)
-          st[SIDX(ST_Z + p + j)] = E[(t - 1 - j) * WAVE + lane];
-        const float ut = U[t * WAVE + lane];
-        for (int i = 0; i < K; ++i) {
-          const float zi = st[SIDX(ST_Z + i)];
-          for (int j = i; j < K; ++j)
-            st[SIDX(ST_A + i * 8 + j)] += zi * st[SIDX(ST_Z + j)];
-          st[SIDX(ST_C + i)] += zi * ut;
-        }
-      }
-      float tr = 0.0f;
-      for (int i = 0; i < K; ++i) tr += st[SIDX(ST_A + i * 8 + i)];
-      const float lam = RIDGE * fmaxf(1.0f, tr / (float)K);
-      for (int i = 0; i < K; ++i) {
-        st[SIDX(ST_A + i * 8 + i)] += lam;
-        for (int j = i + 1; j < K; ++j)
-          st[SIDX(ST_A + j * 8 + i)] = st[SIDX(ST_A + i * 8 + j)];
-      }
-      chol_solve(st, lane, K);
-      float sp = 0.0f, sq = 0.0f;
-      for (int i = 0; i < 4; ++i) PHI(i) = 0.0f;
-      for (int j = 0; j < 4; ++j) THETA(j) = 0.0f;
-      for (int i = 0; i < p; ++i) PHI(i) = st[SIDX(ST_C + i)];
-      for (int j = 0; j < q; ++j) THETA(j) = st[SIDX(ST_C + p + j)];
+  const bool lane_q = q > 0;
+  const bool lane_K = K > 0;
+  int M = 0;
+
+  // ---- stage 2: autocovariance pass + Levinson (q > 0 lanes)
+  if (__any(lane_q)) {
+    M = max(p, q) + 3;
+    if (M > MAXM) M = MAXM;
+    if (M > n / 4) M = max(1, n / 4);
+    float r[MAXM + 1];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) sp += fabsf(PHI(i));
+    for (int k = 0; k <= MAXM; ++k) r[k] = 0.0f;
+    y0 = y1 = 0.0f;
+    for (int t = 0; t < S; ++t) {
+      const float u = u_step(yT, xc, st, lane, g, G, KX, d, t, wm, &y0, &y1);
+      if (t < d) continue;
+      r[0] += u * u;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) sq += fabsf(THETA(j));
-      if (sp > SHRINK) {
-        const float s = SHRINK / sp;
-        for (int i = 0; i < 4; ++i) PHI(i) *= s;
-      }
-      if (sq > SHRINK) {
-        const float s = SHRINK / sq;
-        for (int j = 0; j < 4; ++j) THETA(j) *= s;
-      }
-      // recompute innovations under (phi, theta)
-      const float ph0 = PHI(0), ph1 = PHI(1), ph2 = PHI(2), ph3 = PHI(3);
-      const float th0 = THETA(0), th1 = THETA(1), th2 = THETA(2),
-                  th3 = THETA(3);
-      float e0 = 0.f, e1 = 0.f, e2 = 0.f, e3 = 0.f;
-      for (int t = 0; t < n; ++t) {
-        float acc = U[t * WAVE + lane];
-        if (t - 1 >= 0) acc -= ph0 * U[(t - 1) * WAVE + lane];
-        if (t - 2 >= 0) acc -= ph1 * U[(t - 2) * WAVE + lane];
-        if (t - 3 >= 0) acc -= ph2 * U[(t - 3) * WAVE + lane];
-        if (t - 4 >= 0) acc -= ph3 * U[(t - 4) * WAVE + lane];
-        acc -= th0 * e0 + th1 * e1 + th2 * e2 + th3 * e3;
-        e3 = e2; e2 = e1; e1 = e0; e0 = acc;
-        E[t * WAVE + lane] = acc;
-      }
-      if (q == 0) break;
+      for (int k = 1; k <= MAXM; ++k)
+        if (k <= M) r[k] += u * ur[k - 1];
+      RING_SHIFT(ur, u);
+    }
+    if (lane_q) {
+#pragma unroll
+      for (int k = 0; k <= MAXM; ++k)
+        if (k <= M) st[SIDX(ST_R + k)] = r[k] / (float)n;
+      levinson(st, lane, M);
     }
   }
-  float chk = E[(n - 1) * WAVE + lane] + wm;
+
+  // ---- stages 3+4: two normal-equation passes + final innovation pass.
+  //   NE pass 0: e = long-AR innovations (q>0) / e = u (q==0)
+  //   NE pass 1: e = recursive innovations under pass-0 coefs (q>0 only)
+  //   final pass: e under final coefs -> end-of-series rings (q>0 only;
+  //               q==0 lanes' rings are complete after NE pass 0 and the
+  //               theta terms they feed are zero)
+  const bool any_q = __any(lane_q);
+  if (__any(lane_K)) {
+    const int m = max(max(p, q), 1);
+    // q==0 lanes solve only in pass 0, and a pure-AR wave needs neither
+    // the refinement pass nor the final innovation pass (its theta
+    // terms are zero, so the e-ring is never consumed)
+    const int npass = any_q ? 2 : 1;
+    for (int pass = 0; pass < npass; ++pass) {
+      const bool solve_here = lane_K && (pass == 0 || lane_q);
+      // coefs for this pass's innovation recursion
+      float alr[MAXM], phr[4], thr[4];
+#pragma unroll
+      for (int i = 0; i < MAXM; ++i)
+        alr[i] = (lane_q && pass == 0 && i < M) ? st[SIDX(ST_AL + i)] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) { phr[i] = PHI(i); thr[i] = THETA(i); }
+      if (solve_here) {
+        for (int i = 0; i < 64; ++i) st[SIDX(ST_A + i)] = 0.0f;
+        for (int i = 0; i < 8; ++i) st[SIDX(ST_C + i)] = 0.0f;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { ur[j] = 0.0f; er[j] = 0.0f; }
+      y0 = y1 = 0.0f;
+      for (int t = 0; t < S; ++t) {
+        const float u = u_step(yT, xc, st, lane, g, G, KX, d, t, wm,
+                               &y0, &y1);
+        if (t < d) continue;
+        const int td = t - d;
+        // innovation under this pass's recursion
+        float e = u;
+        if (pass == 0) {
+#pragma unroll
+          for (int i = 0; i < MAXM; ++i) e -= alr[i] * ur[i];
+        } else {
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+            e -= phr[i] * ur[i] + thr[i] * er[i];
+        }
+        if (solve_here && td >= m) {
+#pragma unroll
+          for (int i = 0; i < 4; ++i)
+            if (i < p) st[SIDX(ST_Z + i)] = ur[i];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            if (j < q) st[SIDX(ST_Z + p + j)] = er[j];
+          for (int i = 0; i < K; ++i) {
+            const float zi = st[SIDX(ST_Z + i)];
+            for (int j = i; j < K; ++j)
+              st[SIDX(ST_A + i * 8 + j)] += zi * st[SIDX(ST_Z + j)];
+            st[SIDX(ST_C + i)] += zi * u;
+          }
+        }
+        RING_SHIFT(ur, u);
+        RING_SHIFT(er, e);
+      }
+      if (solve_here) {
+        float tr = 0.0f;
+        for (int i = 0; i < K; ++i) tr += st[SIDX(ST_A + i * 8 + i)];
+        const float lam = RIDGE * fmaxf(1.0f, tr / (float)K);
+        for (int i = 0; i < K; ++i) {
+          st[SIDX(ST_A + i * 8 + i)] += lam;
+          for (int j = i + 1; j < K; ++j)
+            st[SIDX(ST_A + j * 8 + i)] = st[SIDX(ST_A + i * 8 + j)];
+        }
+        chol_solve(st, lane, K);
+        float sp = 0.0f, sq = 0.0f;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) { PHI(i) = 0.0f; THETA(i) = 0.0f; }
+        for (int i = 0; i < p; ++i) PHI(i) = st[SIDX(ST_C + i)];
+        for (int j = 0; j < q; ++j) THETA(j) = st[SIDX(ST_C + p + j)];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) sp += fabsf(PHI(i));
+#pragma unroll
+        for (int j = 0; j < 4; ++j) sq += fabsf(THETA(j));
+        if (sp > SHRINK) {
+          const float s = SHRINK / sp;
+#pragma unroll
+          for (int i = 0; i < 4; ++i) PHI(i) *= s;
+        }
+        if (sq > SHRINK) {
+          const float s = SHRINK / sq;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) THETA(j) *= s;
+        }
+      }
+    }
+    // final innovation pass under the final coefs (leaves the rings at
+    // the end of the series for stage 5 / the fitted-value writer)
+    if (any_q) {
+      float phr[4], thr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) { phr[i] = PHI(i); thr[i] = THETA(i); }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { ur[j] = 0.0f; er[j] = 0.0f; }
+      y0 = y1 = 0.0f;
+      for (int t = 0; t < S; ++t) {
+        const float u = u_step(yT, xc, st, lane, g, G, KX, d, t, wm,
+                               &y0, &y1);
+        if (t < d) continue;
+        float e = u;
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          e -= phr[i] * ur[i] + thr[i] * er[i];
+        RING_SHIFT(ur, u);
+        RING_SHIFT(er, e);
+      }
+    }
+  }
+
+  float chk = ur[0] + er[0] + wm;
 #pragma unroll
   for (int i = 0; i < 4; ++i) chk += PHI(i) + THETA(i);
   for (int k = 0; k < KX; ++k) chk += BETA(k);
@@ -265,15 +376,13 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_eval_kernel(
     const float* __restrict__ w0, const float* __restrict__ w1,
     const float* __restrict__ w2,         // [G] per d, or null
     int T, long long G, int S, int C, int KX) {
-  extern __shared__ float lds[];
-  float* U = lds;
-  float* E = lds + (long long)S * WAVE;
-  float* st = lds + 2 * (long long)S * WAVE;
+  __shared__ float st[NSTATE * WAVE];
   const int lane = threadIdx.x;
   const long long g = (long long)blockIdx.x * WAVE + lane;
   const bool active = g < G;
   const long long gg = active ? g : (G - 1);
   const int H = T - S;
+  float ur[8], er[8];
 
   for (int c = 0; c < C; ++c) {
     const int p = orders[c * 3 + 0];
@@ -286,18 +395,12 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_eval_kernel(
     const float* bpre = (d == 0) ? b0 : (d == 1) ? b1 : b2;
     const float* wpre = (d == 0) ? w0 : (d == 1) ? w1 : w2;
     bool ok;
-    const float wm = fit_lane(yT, xc, PJ, U, E, st, lane, gg, G, S, KX,
-                              p, d, q, &ok, bpre, wpre);
+    const float wm = fit_lane(yT, xc, PJ, st, lane, gg, G, S, KX,
+                              p, d, q, &ok, ur, er, bpre, wpre);
 
-    // ---- stage 5: validation forecast + MSE
-    float u0 = U[(n - 1) * WAVE + lane];
-    float u1 = (n - 2 >= 0) ? U[(n - 2) * WAVE + lane] : 0.f;
-    float u2 = (n - 3 >= 0) ? U[(n - 3) * WAVE + lane] : 0.f;
-    float u3 = (n - 4 >= 0) ? U[(n - 4) * WAVE + lane] : 0.f;
-    float e0 = E[(n - 1) * WAVE + lane];
-    float e1 = (n - 2 >= 0) ? E[(n - 2) * WAVE + lane] : 0.f;
-    float e2 = (n - 3 >= 0) ? E[(n - 3) * WAVE + lane] : 0.f;
-    float e3 = (n - 4 >= 0) ? E[(n - 4) * WAVE + lane] : 0.f;
+    // ---- stage 5: validation forecast + MSE (rings hold u/e at n-1-j)
+    float u0 = ur[0], u1 = ur[1], u2 = ur[2], u3 = ur[3];
+    float e0 = er[0], e1 = er[1], e2 = er[2], e3 = er[3];
     const float ph0 = PHI(0), ph1 = PHI(1), ph2 = PHI(2), ph3 = PHI(3);
     const float th0 = THETA(0), th1 = THETA(1), th2 = THETA(2),
                 th3 = THETA(3);
@@ -327,13 +430,14 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_eval_kernel(
       mse[(long long)c * G + g] = fin ? m : INFINITY;
       statusv[(long long)c * G + g] = fin ? 1 : 0;
     }
-    __syncthreads();  // LDS slabs reused by the next candidate
   }
 }
 
-// Final fit: per-group chosen order (may differ per lane). Writes
-// one-step-ahead fitted values [T][G] in y-units, per-group coefficient
-// vector (wm, beta[KX], phi[4], theta[4]) and status.
+// Final fit: per-group chosen order (may differ per lane — the passes
+// are wave-voted, per-lane work predicated). Writes one-step-ahead
+// fitted values [T][G] in y-units, per-group coefficient vector
+// (wm, beta[KX], phi[4], theta[4]) and status. beta/wm precomputed by
+// the MFMA projection over the FULL-series design when given.
 extern "C" __global__ __launch_bounds__(WAVE) void groupfit_final_kernel(
     const float* __restrict__ yT,
     const float* __restrict__ xc0, const float* __restrict__ xc1,
@@ -344,11 +448,12 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_final_kernel(
     float* __restrict__ fitted,          // [T][G]
     float* __restrict__ params,          // [G][1+KX+8]
     unsigned char* __restrict__ statusv, // [G]
+    const float* __restrict__ b0, const float* __restrict__ b1,
+    const float* __restrict__ b2,        // [KX][G] per d, or null
+    const float* __restrict__ w0, const float* __restrict__ w1,
+    const float* __restrict__ w2,        // [G] per d, or null
     int T, long long G, int KX) {
-  extern __shared__ float lds[];
-  float* U = lds;
-  float* E = lds + (long long)T * WAVE;
-  float* st = lds + 2 * (long long)T * WAVE;
+  __shared__ float st[NSTATE * WAVE];
   const int lane = threadIdx.x;
   const long long g = (long long)blockIdx.x * WAVE + lane;
   const bool active = g < G;
@@ -358,45 +463,57 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_final_kernel(
   const int q = best_order[gg * 3 + 2];
   const float* xc = (d == 0) ? xc0 : (d == 1) ? xc1 : xc2;
   const float* PJ = (d == 0) ? pj0 : (d == 1) ? pj1 : pj2;
+  const float* bpre = (d == 0) ? b0 : (d == 1) ? b1 : b2;
+  const float* wpre = (d == 0) ? w0 : (d == 1) ? w1 : w2;
   const int S = T;  // final fit uses the whole series
-  const int n = S - d;
+  float ur[8], er[8];
 
   bool ok;
-  const float wm = fit_lane(yT, xc, PJ, U, E, st, lane, gg, G, S, KX,
-                            p, d, q, &ok, nullptr, nullptr);
+  const float wm = fit_lane(yT, xc, PJ, st, lane, gg, G, S, KX,
+                            p, d, q, &ok, ur, er, bpre, wpre);
 
-  if (active) {
+  // Fitted-value pass: stream t, predict from the lag rings, THEN push
+  // the current u/e (one-step-ahead: prediction at t uses lags only).
+  {
     const float ph0 = PHI(0), ph1 = PHI(1), ph2 = PHI(2), ph3 = PHI(3);
     const float th0 = THETA(0), th1 = THETA(1), th2 = THETA(2),
                 th3 = THETA(3);
-    float yprev1 = 0.f, yprev2 = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { ur[j] = 0.0f; er[j] = 0.0f; }
+    float y0 = 0.f, y1 = 0.f;
     for (int t = 0; t < T; ++t) {
-      const float yv = yT[(long long)t * G + g];
+      const float yv = yT[(long long)t * G + gg];
+      const float yl1 = y0, yl2 = y1;
+      float w;
+      if (d == 0) w = yv;
+      else if (d == 1) w = yv - y0;
+      else w = yv - 2.0f * y0 + y1;
+      y1 = y0; y0 = yv;
       float out;
       if (t < d) {
         out = yv;
       } else {
         const int td = t - d;
-        float reg = wm;
+        float regx = 0.0f;
         for (int k = 0; k < KX; ++k)
-          reg += xc[(long long)td * KX + k] * BETA(k);
-        float acc = 0.0f;
-        if (td - 1 >= 0) acc += ph0 * U[(td - 1) * WAVE + lane] +
-                                th0 * E[(td - 1) * WAVE + lane];
-        if (td - 2 >= 0) acc += ph1 * U[(td - 2) * WAVE + lane] +
-                                th1 * E[(td - 2) * WAVE + lane];
-        if (td - 3 >= 0) acc += ph2 * U[(td - 3) * WAVE + lane] +
-                                th2 * E[(td - 3) * WAVE + lane];
-        if (td - 4 >= 0) acc += ph3 * U[(td - 4) * WAVE + lane] +
-                                th3 * E[(td - 4) * WAVE + lane];
-        const float what = reg + acc;
+          regx += xc[(long long)td * KX + k] * BETA(k);
+        const float acc = ph0 * ur[0] + ph1 * ur[1] + ph2 * ur[2] +
+                          ph3 * ur[3] + th0 * er[0] + th1 * er[1] +
+                          th2 * er[2] + th3 * er[3];
+        const float what = wm + regx + acc;
         if (d == 0) out = what;
-        else if (d == 1) out = yprev1 + what;
-        else out = 2.0f * yprev1 - yprev2 + what;
+        else if (d == 1) out = yl1 + what;
+        else out = 2.0f * yl1 - yl2 + what;
+        const float u = (w - wm) - regx;
+        const float e = u - acc;
+        RING_SHIFT(ur, u);
+        RING_SHIFT(er, e);
       }
-      fitted[(long long)t * G + g] = out;
-      yprev2 = yprev1; yprev1 = yv;
+      if (active) fitted[(long long)t * G + g] = out;
     }
+  }
+
+  if (active) {
     float* pp = params + g * (1 + KX + 8);
     pp[0] = wm;
     for (int k = 0; k < KX; ++k) pp[1 + k] = BETA(k);
@@ -408,13 +525,6 @@ extern "C" __global__ __launch_bounds__(WAVE) void groupfit_final_kernel(
   }
 }
 
-static void _set_lds_limit(const void* func, size_t shmem) {
-  static size_t done_eval = 0, done_final = 0;
-  (void)done_eval; (void)done_final;
-  hipFuncSetAttribute(func, hipFuncAttributeMaxDynamicSharedMemorySize,
-                      (int)shmem);
-}
-
 extern "C" void launch_groupfit_eval(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
     const float* pj0, const float* pj1, const float* pj2, const int* orders,
@@ -424,10 +534,7 @@ extern "C" void launch_groupfit_eval(
     int T, long long G, int S, int C,
     int KX, hipStream_t stream) {
   const int blocks = (int)((G + WAVE - 1) / WAVE);
-  const size_t shmem =
-      (size_t)(2 * S * WAVE + NSTATE * WAVE) * sizeof(float);
-  _set_lds_limit(reinterpret_cast<const void*>(groupfit_eval_kernel), shmem);
-  hipLaunchKernelGGL(groupfit_eval_kernel, dim3(blocks), dim3(WAVE), shmem,
+  hipLaunchKernelGGL(groupfit_eval_kernel, dim3(blocks), dim3(WAVE), 0,
                      stream, yT, xc0, xc1, xc2, pj0, pj1, pj2, orders, mse,
                      statusv, b0, b1, b2, w0, w1, w2, T, G, S, C, KX);
 }
@@ -436,12 +543,13 @@ extern "C" void launch_groupfit_final(
     const float* yT, const float* xc0, const float* xc1, const float* xc2,
     const float* pj0, const float* pj1, const float* pj2,
     const int* best_order, float* fitted, float* params,
-    unsigned char* statusv, int T, long long G, int KX, hipStream_t stream) {
+    unsigned char* statusv,
+    const float* b0, const float* b1, const float* b2,
+    const float* w0, const float* w1, const float* w2,
+    int T, long long G, int KX, hipStream_t stream) {
   const int blocks = (int)((G + WAVE - 1) / WAVE);
-  const size_t shmem =
-      (size_t)(2 * T * WAVE + NSTATE * WAVE) * sizeof(float);
-  _set_lds_limit(reinterpret_cast<const void*>(groupfit_final_kernel), shmem);
-  hipLaunchKernelGGL(groupfit_final_kernel, dim3(blocks), dim3(WAVE), shmem,
+  hipLaunchKernelGGL(groupfit_final_kernel, dim3(blocks), dim3(WAVE), 0,
                      stream, yT, xc0, xc1, xc2, pj0, pj1, pj2, best_order,
-                     fitted, params, statusv, T, G, KX);
+                     fitted, params, statusv, b0, b1, b2, w0, w1, w2,
+                     T, G, KX);
 }

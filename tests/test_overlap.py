@@ -119,8 +119,11 @@ def test_segmented_allreduce_matches_monolithic():
 @pytest.mark.gpu
 def test_split_capture_matches_monolithic_gpu(monkeypatch):
     """MI355X_GRAPH_OVERLAP=1 forces the two-graph split capture at
-    world_size 1: parameters after k steps must match the monolithic
-    capture bit-for-bit (same kernels, different graph boundaries)."""
+    world_size 1: gradients must match the monolithic capture (lr=0 so
+    warmup cannot move weights; comparing post-Adam params would amplify
+    benign conv-backward noise into lr-scale differences), and repeated
+    replays must be bit-stable — the check that would catch the boundary
+    leaf's .grad accumulating across replays instead of being rewritten."""
     from mi355x_scale.train.graphstep import GraphedTrainStep
 
     dev = torch.device("cuda:0")
@@ -131,23 +134,30 @@ def test_split_capture_matches_monolithic_gpu(monkeypatch):
         "label": torch.randint(0, 10, (8,), generator=g).to(dev),
     }
 
-    def _run(force_overlap: bool):
+    def _grads(force_overlap: bool):
         monkeypatch.setenv("MI355X_GRAPH_OVERLAP",
                            "1" if force_overlap else "0")
         torch.manual_seed(0)
         model = ImageClassifier("resnet18", num_classes=10).to(dev)
-        opt = FlatAdam(model.parameters(), lr=1e-3, bf16_params=True)
+        opt = FlatAdam(model.parameters(), lr=0.0, bf16_params=True)
         gs = GraphedTrainStep(model, opt, batch, world_size=1, warmup=2)
         if force_overlap:
             assert gs.g_bwd2 is not None, "split capture did not engage"
-        for _ in range(4):
-            gs.step(batch)
+        gs.step(batch)
         torch.cuda.synchronize()
-        return opt.flat_master.clone()
+        g1 = torch.cat([b.float().reshape(-1) for b in opt.grad_buffers])
+        g1 = g1.clone()
+        gs.step(batch)  # same weights (lr=0), same batch
+        torch.cuda.synchronize()
+        g2 = torch.cat([b.float().reshape(-1) for b in opt.grad_buffers])
+        assert torch.equal(g1, g2), \
+            "replay not bit-stable: boundary grad accumulated across replays"
+        return g1
 
-    mono = _run(False)
-    split = _run(True)
-    # same kernels, different graph boundaries — identical up to any
-    # conv-backward accumulation nondeterminism
+    mono = _grads(False)
+    split = _grads(True)
+    cos = torch.nn.functional.cosine_similarity(
+        mono.double(), split.double(), dim=0).item()
     rel = ((mono - split).norm() / (mono.norm() + 1e-12)).item()
-    assert rel < 1e-3, f"split-capture params diverged: rel L2 {rel}"
+    assert cos > 0.999, f"gradient cosine {cos}"
+    assert rel < 2e-2, f"gradient relative L2 {rel}"
