@@ -1,0 +1,73 @@
+"""Restartable sharded W1 output (VERDICT next-round #5).
+
+The reference persists per-group forecasts to storage
+(group_apply/02_Fine_Grained_Demand_Forecasting.py:544-552); here the
+forecast frame is written as deterministic group-hash shards with
+atomic renames, so a killed job resumes by skipping finished shards and
+reruns are byte-identical.
+"""
+import os
+
+import pandas as pd
+import pytest
+
+from mi355x_scale.data.generator import generate_demand_data
+from mi355x_scale.forecast import (read_forecast_shards,
+                                   run_fine_grained_forecast_sharded)
+
+
+@pytest.fixture(scope="module")
+def demand():
+    return generate_demand_data(n_products=2, skus_per_product=3,
+                                n_weeks=60)
+
+
+def _run(df, out, **kw):
+    return run_fine_grained_forecast_sharded(
+        df, out, num_shards=4, engine="pandas", horizon=12, max_evals=2,
+        **kw)
+
+
+def test_sharded_write_and_readback(tmp_path, demand):
+    out = str(tmp_path / "fc")
+    paths = _run(demand, out)
+    assert len(paths) == 4
+    assert all(os.path.exists(p) for p in paths)
+    frame = read_forecast_shards(out)
+    # every (Product, SKU, Date) row accounted for exactly once
+    assert len(frame) == len(demand)
+    got = frame.groupby(["Product", "SKU"]).size()
+    want = demand.groupby(["Product", "SKU"]).size()
+    assert got.sort_index().equals(want.sort_index())
+    assert frame["Demand_Fitted"].notna().all()
+
+
+def test_resume_skips_finished_shards_byte_identical(tmp_path, demand):
+    out = str(tmp_path / "fc")
+    paths = _run(demand, out)
+    blobs = {p: open(p, "rb").read() for p in paths}
+    mtimes = {p: os.path.getmtime(p) for p in paths}
+
+    # simulate a crash: one shard lost, one left as a torn tmp file
+    os.unlink(paths[2])
+    with open(paths[1] + ".tmp.999", "wb") as f:
+        f.write(b"torn")
+    _run(demand, out)  # resume
+
+    for p in paths:
+        assert open(p, "rb").read() == blobs[p], f"{p} changed on resume"
+    # finished shards were skipped, not refit (mtime untouched)
+    for p in (paths[0], paths[1], paths[3]):
+        assert os.path.getmtime(p) == mtimes[p]
+    # torn tmp file ignored by the reader
+    frame = read_forecast_shards(out)
+    assert len(frame) == len(demand)
+
+
+def test_rank_sharding_partitions_work(tmp_path, demand):
+    out = str(tmp_path / "fc")
+    _run(demand, out, cur_rank=0, world_size=2)
+    done_r0 = {p for p in os.listdir(out) if p.endswith(".parquet")}
+    assert done_r0 == {"shard-00000.parquet", "shard-00002.parquet"}
+    _run(demand, out, cur_rank=1, world_size=2)
+    assert len(read_forecast_shards(out)) == len(demand)
