@@ -1,0 +1,150 @@
+"""Streaming loader under REAL per-row decode cost (VERDICT r1 #3 /
+reference contract deep_learning/2.distributed-data-loading-petastorm.py:
+282-296,338).
+
+The flagship bench streams pre-decoded bytes (ingest decodes once, a
+documented design choice), so round 1 never measured the loader keeping
+up when the reader pool does actual JPEG decode+resize+crop per row —
+the CPU-bound transform that is the reference's whole stated focus.
+This bench writes an encoded-JPEG dataset and measures:
+
+  1. loader-only rows/sec vs workers_count (where does the pool saturate)
+  2. (GPU) end-to-end train-step samples/sec on the jpeg path vs the
+     raw path's resident ceiling
+
+    python benchmarks/bench_loader_decode.py --rows 4096
+    python benchmarks/bench_loader_decode.py --rows 8192 --train  # GPU
+"""
+import argparse
+import json
+import os
+import sys
+import tempfile
+import time
+
+import numpy as np
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+
+from mi355x_scale.data import BatchReader, DataLoader, DatasetManifest  # noqa: E402
+from mi355x_scale.data.generator import write_jpeg_parquet  # noqa: E402
+
+
+def loader_only(data_dir, workers, batch_size, max_rows,
+                pool_type="thread"):
+    from functools import partial
+
+    from mi355x_scale.data import TransformSpec
+    from mi355x_scale.data.generator import decode_jpeg_batch
+    manifest = DatasetManifest.discover(data_dir)
+    reader = BatchReader(
+        manifest,
+        transform_spec=TransformSpec(
+            partial(decode_jpeg_batch, image_hw=(224, 224)),
+            edit_fields=[("image", np.uint8, (224, 224, 3), False),
+                         ("label", np.int64, (), False)]),
+        workers_count=workers, reader_pool_type=pool_type,
+        results_queue_size=20, num_epochs=None)
+    loader = DataLoader(reader, batch_size)
+    it = iter(loader)
+    # warm: first batch pays thread spin-up
+    next(it)
+    t0 = time.perf_counter()
+    rows = 0
+    while rows < max_rows:
+        b = next(it)
+        rows += len(b["label"])
+    dt = time.perf_counter() - t0
+    loader.close()
+    return rows / dt
+
+
+def train_steps(data_dir, image_format, steps, warmup, batch_size,
+                workers, pool_type="thread"):
+    import torch
+
+    from mi355x_scale.train import ImageClassifier, ImageStreamDataModule
+    from mi355x_scale.train.graphstep import GraphedTrainStep
+    from mi355x_scale.train.flat_adam import FlatAdam
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = ImageClassifier("resnet18", num_classes=1000).to(dev)
+    dm = ImageStreamDataModule(data_dir, batch_size=batch_size,
+                               workers_count=workers,
+                               reader_pool_type=pool_type,
+                               image_format=image_format,
+                               results_queue_size=20)
+    loader = dm.train_dataloader()
+    it = iter(loader)
+    batch = next(it)
+    opt = FlatAdam(model.parameters(), lr=1e-3, bf16_params=True)
+    gs = GraphedTrainStep(model, opt, batch, world_size=1)
+    for _ in range(warmup):
+        gs.step(next(it))
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        gs.step(next(it))
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    dm.teardown()
+    return steps * batch_size / dt
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=4096)
+    ap.add_argument("--batch-size", type=int, default=212)
+    ap.add_argument("--workers-sweep", type=str, default="2,4,8,16,32")
+    ap.add_argument("--train", action="store_true",
+                    help="also run GPU train-step comparisons")
+    ap.add_argument("--train-steps", type=int, default=30)
+    ap.add_argument("--data-dir", type=str, default=None)
+    args = ap.parse_args()
+
+    d = args.data_dir or os.path.join(tempfile.gettempdir(),
+                                      f"jpegds_{args.rows}")
+    if not os.path.exists(os.path.join(d, "part-00000.parquet")):
+        t0 = time.perf_counter()
+        write_jpeg_parquet(d, num_rows=args.rows, rows_per_group=64)
+        print(f"# wrote {args.rows} jpeg rows in "
+              f"{time.perf_counter() - t0:.1f}s -> {d}", file=sys.stderr)
+
+    out = {"metric": "loader rows/sec (jpeg decode path)",
+           "rows": args.rows, "sweep": {}}
+    for pool in ("thread", "process"):
+        out["sweep"][pool] = {}
+        for wstr in args.workers_sweep.split(","):
+            wk = int(wstr)
+            rps = loader_only(d, wk, args.batch_size,
+                              max_rows=min(args.rows * 2, 6000),
+                              pool_type=pool)
+            out["sweep"][pool][wk] = round(rps, 1)
+            print(f"# pool={pool} workers={wk}: {rps:,.0f} rows/s",
+                  file=sys.stderr)
+
+    if args.train:
+        import torch
+        assert torch.cuda.is_available()
+        from mi355x_scale.data.generator import write_image_parquet
+        draw = os.path.join(tempfile.gettempdir(), "rawds_bench")
+        if not os.path.exists(os.path.join(draw, "part-00000.parquet")):
+            write_image_parquet(draw, num_rows=4096, rows_per_group=212,
+                                rows_per_file=2120)
+        best_pool, best_workers, best_rps = max(
+            ((pool, wk, rps) for pool, sw in out["sweep"].items()
+             for wk, rps in sw.items()), key=lambda t: t[2])
+        jp = train_steps(d, "jpeg", args.train_steps, 10,
+                         args.batch_size, best_workers,
+                         pool_type=best_pool)
+        raw = train_steps(draw, "raw", args.train_steps, 10,
+                          args.batch_size, 2)
+        out["train_jpeg_samples_per_s"] = round(jp, 1)
+        out["train_raw_samples_per_s"] = round(raw, 1)
+        out["jpeg_pool"] = best_pool
+        out["jpeg_workers"] = best_workers
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

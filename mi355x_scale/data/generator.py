@@ -225,6 +225,91 @@ def write_image_parquet(
     return out_dir
 
 
+def write_jpeg_parquet(
+    out_dir: str,
+    num_rows: int = 2048,
+    source_hw_range: Tuple[int, int] = (256, 384),
+    num_classes: int = 1000,
+    rows_per_group: int = 64,
+    rows_per_file: int = 512,
+    quality: int = 85,
+    seed: int = SEED,
+) -> str:
+    """Write a synthetic dataset of ENCODED JPEG bytes (variable-size
+    binary column) + label — the reference's actual Delta schema (raw
+    ``content`` bytes from ``binaryFile``,
+    ``deep_learning/1.data-preparation.py:118-124``), for exercising the
+    loader under the real per-row decode cost the reference's
+    TransformSpec pays every epoch (``deep_learning/2...py:282-296``).
+    Images are smooth gradients + noise at varying source sizes so JPEG
+    encode/decode cost and the resize path are realistic.
+    """
+    import io
+
+    from PIL import Image
+
+    os.makedirs(out_dir, exist_ok=True)
+    rng = np.random.default_rng(seed)
+    file_idx = 0
+    written = 0
+    lo, hi = source_hw_range
+    while written < num_rows:
+        n = min(rows_per_file, num_rows - written)
+        blobs = []
+        for _ in range(n):
+            h = int(rng.integers(lo, hi + 1))
+            w = int(rng.integers(lo, hi + 1))
+            yy, xx = np.mgrid[0:h, 0:w]
+            base = np.stack([
+                (127 + 120 * np.sin(xx / rng.uniform(8, 40))),
+                (127 + 120 * np.cos(yy / rng.uniform(8, 40))),
+                (127 + 120 * np.sin((xx + yy) / rng.uniform(8, 40))),
+            ], axis=-1)
+            img = np.clip(base + rng.normal(0, 12, size=(h, w, 3)),
+                          0, 255).astype(np.uint8)
+            buf = io.BytesIO()
+            Image.fromarray(img).save(buf, "JPEG", quality=quality)
+            blobs.append(buf.getvalue())
+        labels = rng.integers(0, num_classes, size=n, dtype=np.int64)
+        table = pa.table({
+            "image": pa.array(blobs, type=pa.binary()),
+            "label": pa.array(labels),
+        })
+        path = os.path.join(out_dir, f"part-{file_idx:05d}.parquet")
+        pq.write_table(table, path, row_group_size=rows_per_group,
+                       compression="none")
+        file_idx += 1
+        written += n
+    return out_dir
+
+
+def decode_jpeg_batch(pdf: pd.DataFrame,
+                      image_hw: Tuple[int, int] = (224, 224)
+                      ) -> Dict[str, np.ndarray]:
+    """Reference-style per-row CPU transform (``deep_learning/2...py:
+    282-296``): JPEG decode → resize(short side 256-scaled) →
+    center-crop. Output stays uint8 NHWC — normalization/CHW runs on
+    device in the fused HIP kernel, where the reference burned CPU."""
+    import io
+
+    from PIL import Image
+
+    h, w = image_hw
+    out = np.empty((len(pdf), h, w, 3), dtype=np.uint8)
+    for i, b in enumerate(pdf["image"]):
+        with Image.open(io.BytesIO(b)) as im:
+            im = im.convert("RGB")
+            scale = max(h / im.height, w / im.width) * 256.0 / 224.0
+            nh = int(round(im.height * scale))
+            nw = int(round(im.width * scale))
+            im = im.resize((nw, nh))
+            left = (nw - w) // 2
+            top = (nh - h) // 2
+            out[i] = np.asarray(im.crop((left, top, left + w, top + h)),
+                                dtype=np.uint8)
+    return {"image": out, "label": pdf["label"].to_numpy()}
+
+
 def decode_image_batch(pdf: pd.DataFrame, image_hw: Tuple[int, int] = (224, 224)) -> Dict[str, np.ndarray]:
     """Default transform for the synthetic image dataset: bytes → uint8
     NHWC array. Normalization/CHW happens on-device (fused HIP kernel) —

@@ -38,6 +38,17 @@ from .transform import TransformSpec
 _STOP = object()
 
 
+class _SpecAsArrowTransform:
+    """Adapts a TransformSpec to the process pool's table→dict contract
+    (picklable; runs the pandas round-trip inside the worker process)."""
+
+    def __init__(self, spec: TransformSpec):
+        self.spec = spec
+
+    def __call__(self, table):
+        return self.spec.apply(table.to_pandas())
+
+
 class BatchReader:
     """Iterator over decoded row-group batches (dict[str, np.ndarray])."""
 
@@ -102,16 +113,17 @@ class BatchReader:
                 t.start()
                 self._threads.append(t)
         elif reader_pool_type == "process":
-            if transform_spec is not None:
-                raise ValueError(
-                    "process pool supports arrow_transform (picklable) "
-                    "only; use reader_pool_type='thread' for "
-                    "transform_spec")
             from .process_pool import ProcessReaderPool
+            at = arrow_transform
+            if transform_spec is not None:
+                # GIL-heavy user transforms (e.g. per-row JPEG decode)
+                # are exactly what the process pool is for: ship the
+                # spec via cloudpickle, apply to the decoded pandas
+                # batch inside the worker process
+                at = _SpecAsArrowTransform(transform_spec)
             self._proc_pool = ProcessReaderPool(
                 self._refs, self.workers_count, num_epochs,
-                schema_fields, arrow_transform,
-                results_queue_size)
+                schema_fields, at, results_queue_size)
 
     # -- work distribution ------------------------------------------------
     def _next_ref(self) -> Optional[RowGroupRef]:

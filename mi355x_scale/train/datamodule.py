@@ -70,7 +70,11 @@ class ImageStreamDataModule(DataModule):
         prefetch_depth: int = 2,
         stagers: int = 2,
         val_fraction_shards: bool = True,
+        image_format: str = "raw",
     ):
+        if image_format not in ("raw", "jpeg"):
+            raise ValueError("image_format must be 'raw' or 'jpeg'")
+        self.image_format = image_format
         self.data_dir = data_dir
         self.batch_size = batch_size
         self.workers_count = workers_count
@@ -99,15 +103,29 @@ class ImageStreamDataModule(DataModule):
     def _make_loader(self, num_epochs: Optional[int]):
         self.setup()
         h, w = self.image_hw
+        if self.image_format == "jpeg":
+            # encoded-JPEG datasets: the reference's real per-row CPU
+            # transform (decode + resize + crop) runs in the reader pool
+            # (deep_learning/2...py:282-296); normalize stays on device
+            from functools import partial
+
+            from ..data.generator import decode_jpeg_batch
+            kwargs = dict(transform_spec=TransformSpec(
+                partial(decode_jpeg_batch, image_hw=(h, w)),
+                edit_fields=[("image", np.uint8, (h, w, 3), False),
+                             ("label", np.int64, (), False)],
+                selected_fields=["image", "label"]))
+        else:
+            kwargs = dict(arrow_transform=_arrow_images(h, w))
         reader = BatchReader(
             self._manifest,
-            arrow_transform=_arrow_images(h, w),
             cur_shard=self.cur_shard,
             shard_count=self.shard_count,
             workers_count=self.workers_count,
             reader_pool_type=self.reader_pool_type,
             results_queue_size=self.results_queue_size,
             num_epochs=num_epochs,
+            **kwargs,
         )
         loader = DeviceLoader(DataLoader(reader, self.batch_size),
                               self.device, depth=self.prefetch_depth,
