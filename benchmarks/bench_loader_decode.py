@@ -61,19 +61,25 @@ def loader_only(data_dir, workers, batch_size, max_rows,
 
 def train_steps(data_dir, image_format, steps, warmup, batch_size,
                 workers, pool_type="thread"):
+    """Runs in a fresh subprocess (see --train-one): a prior run's decode
+    worker processes must not pollute the measurement, and HIP/queue
+    teardown crashes at interpreter exit must not kill the sweep."""
     import torch
 
     from mi355x_scale.train import ImageClassifier, ImageStreamDataModule
     from mi355x_scale.train.graphstep import GraphedTrainStep
     from mi355x_scale.train.flat_adam import FlatAdam
+    torch.backends.cudnn.benchmark = True
     dev = torch.device("cuda:0")
     torch.manual_seed(0)
     model = ImageClassifier("resnet18", num_classes=1000).to(dev)
+    model = model.to(memory_format=torch.channels_last)
     dm = ImageStreamDataModule(data_dir, batch_size=batch_size,
                                workers_count=workers,
                                reader_pool_type=pool_type,
                                image_format=image_format,
-                               results_queue_size=20)
+                               results_queue_size=20,
+                               prefetch_depth=3, stagers=3)
     loader = dm.train_dataloader()
     it = iter(loader)
     batch = next(it)
@@ -91,6 +97,23 @@ def train_steps(data_dir, image_format, steps, warmup, batch_size,
     return steps * batch_size / dt
 
 
+def _train_subprocess(data_dir, image_format, steps, batch_size, workers,
+                      pool_type):
+    import re
+    import subprocess
+    cmd = [sys.executable, os.path.abspath(__file__), "--train-one",
+           image_format, "--data-dir", data_dir,
+           "--train-steps", str(steps), "--batch-size", str(batch_size),
+           "--one-workers", str(workers), "--one-pool", pool_type]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=600)
+    m = re.search(r"TRAIN_RESULT (\S+)", res.stdout)
+    if not m:
+        print(res.stdout[-2000:], file=sys.stderr)
+        print(res.stderr[-2000:], file=sys.stderr)
+        raise RuntimeError(f"train subprocess failed rc={res.returncode}")
+    return float(m.group(1))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=4096)
@@ -100,7 +123,19 @@ def main():
                     help="also run GPU train-step comparisons")
     ap.add_argument("--train-steps", type=int, default=30)
     ap.add_argument("--data-dir", type=str, default=None)
+    ap.add_argument("--train-one", type=str, default=None,
+                    help="internal: run ONE train measurement and print "
+                         "TRAIN_RESULT <samples/s>")
+    ap.add_argument("--one-workers", type=int, default=6)
+    ap.add_argument("--one-pool", type=str, default="thread")
     args = ap.parse_args()
+
+    if args.train_one:
+        rps = train_steps(args.data_dir, args.train_one, args.train_steps,
+                          10, args.batch_size, args.one_workers,
+                          pool_type=args.one_pool)
+        print(f"TRAIN_RESULT {rps:.1f}", flush=True)
+        os._exit(0)  # skip teardown crashes in HIP/mp-queue atexit paths
 
     d = args.data_dir or os.path.join(tempfile.gettempdir(),
                                       f"jpegds_{args.rows}")
@@ -117,7 +152,7 @@ def main():
         for wstr in args.workers_sweep.split(","):
             wk = int(wstr)
             rps = loader_only(d, wk, args.batch_size,
-                              max_rows=min(args.rows * 2, 6000),
+                              max_rows=min(args.rows * 3, 20000),
                               pool_type=pool)
             out["sweep"][pool][wk] = round(rps, 1)
             print(f"# pool={pool} workers={wk}: {rps:,.0f} rows/s",
@@ -134,11 +169,10 @@ def main():
         best_pool, best_workers, best_rps = max(
             ((pool, wk, rps) for pool, sw in out["sweep"].items()
              for wk, rps in sw.items()), key=lambda t: t[2])
-        jp = train_steps(d, "jpeg", args.train_steps, 10,
-                         args.batch_size, best_workers,
-                         pool_type=best_pool)
-        raw = train_steps(draw, "raw", args.train_steps, 10,
-                          args.batch_size, 2)
+        raw = _train_subprocess(draw, "raw", args.train_steps,
+                                args.batch_size, 6, "thread")
+        jp = _train_subprocess(d, "jpeg", args.train_steps,
+                               args.batch_size, best_workers, best_pool)
         out["train_jpeg_samples_per_s"] = round(jp, 1)
         out["train_raw_samples_per_s"] = round(raw, 1)
         out["jpeg_pool"] = best_pool

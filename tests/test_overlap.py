@@ -150,8 +150,15 @@ def test_split_capture_matches_monolithic_gpu(monkeypatch):
         gs.step(batch)  # same weights (lr=0), same batch
         torch.cuda.synchronize()
         g2 = torch.cat([b.float().reshape(-1) for b in opt.grad_buffers])
-        assert torch.equal(g1, g2), \
-            "replay not bit-stable: boundary grad accumulated across replays"
+        # MIOpen's atomic-accumulation wrw kernels are not bit-stable
+        # across replays (even monolithic), but an accumulating boundary
+        # grad would DOUBLE the early-layer grads on the second replay —
+        # catch that by norm ratio, tolerate atomics noise by rel diff
+        ratio = (g2.norm() / (g1.norm() + 1e-12)).item()
+        rel = ((g2 - g1).norm() / (g1.norm() + 1e-12)).item()
+        assert 0.95 < ratio < 1.05, \
+            f"grad norm grew {ratio}x across replays (accumulation bug)"
+        assert rel < 1e-2, f"replays diverged: rel L2 {rel}"
         return g1
 
     mono = _grads(False)
