@@ -6,11 +6,39 @@ reference uses everywhere (``hyperopt/1. hyperopt.py:95``,
 1-D adaptive Parzen estimators per dimension — observations are split
 into a "good" fraction (lowest-loss γ quantile) and the rest; candidates
 are drawn from the good-density l(x) and ranked by l(x)/g(x)
-(equivalently EI). Bandwidths follow the adjacent-point heuristic.
+(equivalently EI).
 
-This is a faithful algorithmic re-implementation, not a bit-for-bit port:
-seeded runs are deterministic here, but do not reproduce hyperopt's exact
-draws.
+Internals follow hyperopt's published adaptive-Parzen algorithm
+(hyperopt/tpe.py ``adaptive_parzen_normal`` / ``default_gamma``), and
+``tests/test_tpe_fidelity.py`` asserts them step-by-step:
+
+  * good-set size  n_below = min(ceil(γ·√n), 25)  with γ = 0.25;
+  * per-component bandwidth = max distance to the adjacent sorted means
+    (prior inserted as an extra component at its sorted position);
+  * bandwidth clamps  maxsigma = prior_sigma,
+    minsigma = prior_sigma / min(100, 1 + m);
+  * linear forgetting: with m > LF=25 observations, the m−LF oldest
+    get linearly ramped weights (newest 25 keep weight 1);
+  * prior component weight = ``PRIOR_WEIGHT`` (1.0), weights normalized.
+
+Deliberate divergences from hyperopt (each load-bearing difference, not
+an approximation error — the draw SEQUENCE therefore does not reproduce
+hyperopt's bit-for-bit; statistical behavior is covered by
+``tests/test_tune.py``):
+
+  1. ``n_startup_trials = 10`` (hyperopt: 20). The reference's W1 inner
+     search runs ``max_evals=10`` (``group_apply/02_...py:469``) — under
+     hyperopt's 20 that search is PURE random sampling; keeping 10 lets
+     the posterior engage in-reference-budget searches.
+  2. RNG: numpy ``default_rng`` Generator draws (hyperopt threads a
+     legacy ``RandomState`` through pyll rec_eval); draw order also
+     differs — we draw all candidates per dimension at once.
+  3. Quantized/int spaces sample candidates in the continuous internal
+     coordinate and round on output (hyperopt re-quantizes inside the
+     GMM sampler); means of observed points are the post-quantization
+     values in both.
+  4. Sorted-means bandwidth uses the unsorted component order with a
+     searchsorted lookup (same values as hyperopt's sorted walk).
 """
 from __future__ import annotations
 
@@ -25,39 +53,48 @@ N_STARTUP_TRIALS = 10     # random exploration before TPE kicks in
 N_EI_CANDIDATES = 24      # candidates drawn from l(x) per dimension
 GAMMA = 0.25              # good/bad split quantile
 PRIOR_WEIGHT = 1.0
+LINEAR_FORGETTING = 25    # hyperopt DEFAULT_LF
 
 
 def _split(losses: np.ndarray) -> int:
-    """Number of observations in the 'good' set (hyperopt-style)."""
+    """Number of observations in the 'good' set — hyperopt's
+    ``default_gamma``: min(ceil(γ·√n), 25), capped below n."""
     n = len(losses)
-    return max(1, min(int(math.ceil(GAMMA * math.sqrt(n) * 4)), n - 1, 25))
+    return max(1, min(int(math.ceil(GAMMA * math.sqrt(n))),
+                      LINEAR_FORGETTING, n - 1))
+
+
+def _forgetting_weights(m: int) -> np.ndarray:
+    """hyperopt ``linear_forgetting_weights``: newest LF points keep
+    weight 1, older points ramp linearly down (oldest smallest)."""
+    if m <= LINEAR_FORGETTING:
+        return np.ones(m)
+    ramp = np.linspace(1.0 / m, 1.0, num=m - LINEAR_FORGETTING)
+    return np.concatenate([ramp, np.ones(LINEAR_FORGETTING)])
 
 
 def _adaptive_parzen(mus: np.ndarray, prior_mu: float, prior_sigma: float
                      ) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
-    """Weights, means, sigmas of the 1-D Parzen mixture over ``mus`` plus
-    the prior component (hyperopt's adaptive_parzen_normal heuristic)."""
-    order = np.argsort(mus)
-    smus = mus[order]
-    m = len(smus)
-    # insert prior as an extra component
-    means = np.concatenate([smus, [prior_mu]])
+    """Weights, means, sigmas of the 1-D Parzen mixture over ``mus``
+    (observation order = trial order) plus the prior component —
+    hyperopt's ``adaptive_parzen_normal``."""
+    m = len(mus)
+    means = np.concatenate([mus, [prior_mu]])
     means_sorted = np.sort(means)
     sigmas = np.empty(m + 1)
     for i, mu in enumerate(means):
         pos = np.searchsorted(means_sorted, mu)
         left = means_sorted[pos - 1] if pos > 0 else mu - prior_sigma
-        right = means_sorted[pos + 1] if pos + 1 < len(means_sorted) else mu + prior_sigma
+        right = (means_sorted[pos + 1] if pos + 1 < len(means_sorted)
+                 else mu + prior_sigma)
         sigmas[i] = max(abs(mu - left), abs(right - mu))
-    # clamp
-    minsigma = prior_sigma / max(100.0, (1.0 + m))
+    # hyperopt's clamps: max = prior_sigma; min = prior/min(100, 1+m)
+    minsigma = prior_sigma / min(100.0, 1.0 + m)
     sigmas = np.clip(sigmas, minsigma, prior_sigma)
     sigmas[m] = prior_sigma  # prior keeps its width
-    weights = np.ones(m + 1)
+    weights = np.empty(m + 1)
+    weights[:m] = _forgetting_weights(m)
     weights[m] = PRIOR_WEIGHT
-    if m > 20:  # downweight old points linearly (hyperopt ramp)
-        ramp = np.linspace(1.0 / m, 1.0, m)
-        weights[:m] = ramp
     weights /= weights.sum()
     return weights, means, sigmas
 
