@@ -9,8 +9,16 @@ implementations run (they are also the numerics references in tests).
 from __future__ import annotations
 
 
+import os as _os
+
 try:
-    from . import _C  # type: ignore
+    if _os.environ.get("MI355X_OPS_EXT") == "_C_asan":
+        # SURVEY §5.2 sanitizer pass: the AddressSanitizer device build
+        # (setup.py MI355X_ASAN=1) substitutes for _C so the whole GPU
+        # kernel test suite runs instrumented (tools/gpu_sanitize.sh)
+        from . import _C_asan as _C  # type: ignore
+    else:
+        from . import _C  # type: ignore
     HAVE_EXT = True
 except ImportError:
     _C = None

@@ -24,11 +24,34 @@ SRC = [
     "mi355x_scale/ops/csrc/arma_gen.hip",
 ]
 
+ASAN = os.environ.get("MI355X_ASAN") == "1"
+
+if ASAN:
+    # device AddressSanitizer build of the kernel extension
+    # (gfx950:xnack+ + asanrtl.bc device runtime). Built as a separate
+    # module; tools/gpu_sanitize.py drives every kernel through it once
+    # per round on a GPU box under HSA_XNACK=1.
+    asan_flags = ["-g", "-O1", "-fsanitize=address",
+                  "-shared-libsan", "--offload-arch=gfx950:xnack+"]
+    ext_modules = [
+        CUDAExtension(
+            name="mi355x_scale.ops._C_asan",
+            sources=SRC,
+            extra_compile_args={"cxx": ["-g", "-O1", "-fsanitize=address"],
+                                "nvcc": asan_flags},
+            # the final link runs through g++ (no -shared-libsan there);
+            # -fsanitize=address pulls the asan runtime it knows
+            extra_link_args=["-fsanitize=address"],
+        ),
+    ]
+else:
+    ext_modules = None  # filled below
+
 setup(
     name="mi355x_scale",
     version="0.1.0",
     packages=["mi355x_scale"],
-    ext_modules=[
+    ext_modules=ext_modules if ASAN else [
         CUDAExtension(
             name="mi355x_scale.ops._C",
             sources=SRC,
