@@ -19,6 +19,7 @@ import torch.nn as nn
 
 from ..ops.fused_bn import FusedBNReLU2d
 from ..ops.maxpool import MaxPool3x3s2
+from ..ops.stemconv import StemConv2d
 
 
 def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -74,7 +75,7 @@ class ResNet(nn.Module):
                  zero_init_residual: bool = True):
         super().__init__()
         self.inplanes = 64
-        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.conv1 = StemConv2d()  # hand-written MFMA 7x7 s2 (ops/csrc/stemconv.hip)
         self.bn1 = FusedBNReLU2d(64)  # stem BN+ReLU in one pass
         self.maxpool = MaxPool3x3s2()  # u8-code NHWC pool kernel on GPU
         self.layer1 = self._make_layer(block, 64, layers[0])

@@ -60,6 +60,14 @@ extern "C" void launch_groupfit_final(
     const float* w0, const float* w1, const float* w2,
     int T, long long G, int KX, hipStream_t stream);
 
+extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
+                                     int Nb, int H, int W, int HO, int WO,
+                                     hipStream_t stream, int phase_mask);
+extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
+                                     float* dw_f32, void* dw_bf16,
+                                     int Nb, int H, int W, int HO, int WO,
+                                     hipStream_t stream);
+
 static void _check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
                   t.is_contiguous(),
@@ -455,7 +463,50 @@ void adam_step_mixed(torch::Tensor master, torch::Tensor gb,
                          at::cuda::getCurrentHIPStream().stream());
 }
 
+// Stem conv (7x7 s2 p3, 3->64, NHWC bf16): out = conv(x, w).
+// x [N,H,W,3] bf16 channels-last storage, w [64,3,7,7] channels_last,
+// out [N,HO,WO,64].
+void stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
+                   int64_t phase_mask = 7) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(out.is_cuda() && out.scalar_type() == torch::kBFloat16);
+  int Nb = x.size(0), H = x.size(1), W = x.size(2);
+  TORCH_CHECK(x.size(3) == 3 && out.size(3) == 64, "stem is 3->64");
+  int HO = out.size(1), WO = out.size(2);
+  TORCH_CHECK(WO <= 128, "stem kernel handles output rows up to 128 px");
+  TORCH_CHECK(w.numel() == 64 * 7 * 7 * 3);
+  launch_stem_conv_fwd(x.data_ptr(), w.data_ptr(), out.data_ptr(),
+                       Nb, H, W, HO, WO,
+                       at::cuda::getCurrentHIPStream().stream(),
+                       (int)phase_mask);
+}
+
+// Weight grad: dw (bf16, weight layout) from x and dy; dw_f32 is the
+// [64][160] fp32 accumulation scratch.
+void stem_conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dw_f32,
+                   torch::Tensor dw) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(dw_f32.scalar_type() == torch::kFloat32 &&
+              dw_f32.numel() == 64 * 160);
+  TORCH_CHECK(dw.scalar_type() == torch::kBFloat16 &&
+              dw.numel() == 64 * 7 * 7 * 3);
+  int Nb = x.size(0), H = x.size(1), W = x.size(2);
+  int HO = dy.size(1), WO = dy.size(2);
+  launch_stem_conv_wrw(x.data_ptr(), dy.data_ptr(),
+                       dw_f32.data_ptr<float>(), dw.data_ptr(),
+                       Nb, H, W, HO, WO,
+                       at::cuda::getCurrentHIPStream().stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("stem_conv_fwd", &stem_conv_fwd,
+        "MFMA stem conv fwd (7x7 s2, 3->64, NHWC bf16)",
+        pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("out"),
+        pybind11::arg("phase_mask") = 7);
+  m.def("stem_conv_wrw", &stem_conv_wrw,
+        "MFMA stem conv weight-grad (fp32 accum + bf16 cast)");
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
   m.def("groupfit_eval", &groupfit_eval,

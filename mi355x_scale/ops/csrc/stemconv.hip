@@ -1,0 +1,304 @@
+// Hand-written MFMA stem convolution for gfx950: 7x7 stride-2 pad-3,
+// Cin=3, Cout=64, NHWC bf16 (fwd + weight-grad; the stem input never
+// needs a data-grad).
+//
+// Why: the ResNet stem is the one conv MIOpen leaves on a generic igemm
+// tile (C=3 starves the GEMM K: K = 7*7*3 = 147) — measured 298 us fwd
+// + ~294 us wrw per step at bs 212, ~7% of the whole flagship step
+// (profiles/r01_bench_resnet18_fused_kernstats.md). A shape-special
+// kernel treats it as the GEMM it is:
+//
+//   fwd:  out[px][co]  = sum_k  patch[px][k]   * w[co][k]     (M=N*HO*WO, N=64, K=147)
+//   wrw:  dw[co][k]    = sum_px dy[px][co]     * patch[px][k] (M=64, N=147, K=N*HO*WO)
+//
+// on v_mfma_f32_16x16x32_bf16 (frag maps per cdna_hip_programming.md §3:
+// A row = lane&15, B col = lane&15, k-slice = 8*(lane>>4)+j; C/D
+// col = lane&15, row = 4*(lane>>4)+reg).
+//
+// Memory discipline (ablated, v1/v2 lessons):
+//   v1: per-element GLOBAL im2col gather — one load+wait per element,
+//       waves >80% stalled (SQ_WAIT_ANY), 2.4 ms.
+//   v2: contiguous band stage + LDS->LDS im2col expansion — the MFMA
+//       phase alone measured 108 us, but stage (637 us) and expand
+//       (541 us) still dominated: serialized load->ds_write round trips.
+//   v3 (this): each block owns ONE output row (n, ho); the 7-row input
+//       band is staged with short4 vector copies into a PIXEL-PADDED
+//       layout (edge zeros materialized, plus an all-zero 8th row that
+//       absorbs the K-padding taps), so there is no expansion phase and
+//       no bounds checks: the MFMA loop gathers its A-fragments straight
+//       from the band at koff(k) + px*6.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define KH 7
+#define KW 7
+#define CI 3
+#define CO 64
+#define KTAP (KH * KW * CI)     // 147
+#define KPAD 160                // 5 x K32 MFMA steps
+#define KLDS 168                // padded weight-tile row stride
+#define PXPAD 128               // padded output-row length (WO <= 128)
+// band row: 4 zero pixels of lead (data starts 8B-aligned at el 12),
+// 2*PXPAD+KW pixel taps, zero tail; rounded to short4 granularity
+#define XTROW ((4 + 2 * PXPAD + KW + 4) * CI + 3)  // 816 els
+#define XTOFF 12                // data starts at pixel 4 -> element 12
+#define STRIDE 2
+#define PADDING 3
+
+typedef __bf16 bf16;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// Zero the whole 8-row band, then copy the 7 valid rows (short4 vector
+// copies when the source rows are 8B-aligned). Callers barrier before
+// the MFMA phase.
+__device__ __forceinline__ void stage_band(
+    const bf16* __restrict__ x, bf16* xt, int n, int ho, int H, int W,
+    int tid, int nthreads) {
+  for (int i = tid; i < 8 * XTROW; i += nthreads) xt[i] = (bf16)0.0f;
+  __syncthreads();
+  const int rowlen = W * CI;
+  const bool vec_ok = (rowlen & 3) == 0;
+  for (int r = 0; r < KH; ++r) {
+    const int ih = ho * STRIDE - PADDING + r;
+    if (ih < 0 || ih >= H) continue;
+    const bf16* src = x + ((long long)n * H + ih) * rowlen;
+    bf16* dst = xt + r * XTROW + XTOFF;
+    if (vec_ok) {
+      const bf16x4* s4 = reinterpret_cast<const bf16x4*>(src);
+      bf16x4* d4 = reinterpret_cast<bf16x4*>(dst);
+      const int nch = rowlen >> 2;
+      for (int i = tid; i < nch; i += nthreads) d4[i] = s4[i];
+    } else {
+      for (int i = tid; i < rowlen; i += nthreads) dst[i] = src[i];
+    }
+  }
+}
+
+// koff(k): element offset of tap k for pixel 0 in the padded band.
+// tap iw = px*STRIDE - PADDING + kw -> element
+// XTOFF + (px*STRIDE - PADDING + kw)*CI + ci
+//   = (XTOFF - PADDING*CI) + px*(STRIDE*CI) + (kw*CI + ci)
+// pad taps (k >= 147) resolve to kh = 7, the all-zero row.
+__device__ __forceinline__ int koff_of(int k) {
+  const int kh = k / (KW * CI);
+  const int r21 = k - kh * (KW * CI);
+  return kh * XTROW + (XTOFF - PADDING * CI) + r21;
+}
+
+// ---------------------------------------------------------------- forward
+// block = 256 threads (4 waves) = one output row; wave w computes px
+// subtiles {2w, 2w+1}. LDS = band (12.8 KB) + weight tile (21 KB) ->
+// 4 blocks/CU resident.
+extern "C" __global__ __launch_bounds__(256) void stem_conv_fwd_kernel(
+    const bf16* __restrict__ x,   // [N][H][W][CI]
+    const bf16* __restrict__ w,   // [CO][KH][KW][CI] (channels_last)
+    bf16* __restrict__ out,       // [N][HO][WO][CO]
+    int Nb, int H, int W, int HO, int WO, int phase_mask) {
+  __shared__ bf16 xt[8 * XTROW];
+  __shared__ bf16 Wl[CO * KLDS];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int n = blockIdx.x / HO;
+  const int ho = blockIdx.x - n * HO;
+
+  if (phase_mask & 1) {
+    // weight tile: 40 els/thread, loads batched 8 deep
+    for (int base = tid * 8; base < CO * KPAD; base += 256 * 8) {
+      bf16 v[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int idx = base + u;
+        const int co = idx / KPAD, k = idx - co * KPAD;
+        v[u] = (idx < CO * KPAD && k < KTAP) ? w[co * KTAP + k]
+                                             : (bf16)0.0f;
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int idx = base + u;
+        if (idx < CO * KPAD) {
+          const int co = idx / KPAD, k = idx - co * KPAD;
+          Wl[co * KLDS + k] = v[u];
+        }
+      }
+    }
+    stage_band(x, xt, n, ho, H, W, tid, 256);
+  }
+  __syncthreads();
+  if (!(phase_mask & 4)) return;
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int c = 0; c < 4; ++c) acc[m][c] = (f32x4)0.0f;
+
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int pxb0 = (wave * 32 + row16) * (STRIDE * CI);
+  const int pxb1 = (wave * 32 + 16 + row16) * (STRIDE * CI);
+#pragma unroll
+  for (int kk = 0; kk < KPAD / 32; ++kk) {
+    bf16x8 bfrag[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      bfrag[c] = *reinterpret_cast<const bf16x8*>(
+          Wl + (c * 16 + row16) * KLDS + kk * 32 + kgrp * 8);
+    bf16x8 f0, f1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int ko = koff_of(kk * 32 + kgrp * 8 + j);
+      f0[j] = xt[ko + pxb0];
+      f1[j] = xt[ko + pxb1];
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      acc[0][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          f0, bfrag[c], acc[0][c], 0, 0, 0);
+      acc[1][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          f1, bfrag[c], acc[1][c], 0, 0, 0);
+    }
+  }
+
+  // store: D col = lane&15 (co), row = 4*(lane>>4)+r (px within row)
+  const long long rowbase = ((long long)n * HO + ho) * WO;
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    const int pxr = wave * 32 + m * 16 + kgrp * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int px = pxr + r;
+      if (px >= WO) continue;
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        out[(rowbase + px) * CO + c * 16 + row16] = (bf16)acc[m][c][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------- wrw
+// block = 256 threads (4 waves) loops over ``rows_per_block`` output
+// rows; per row: band stage + dy-row stage, then each wave reduces ALL
+// four 32-px slices into its own (co, k) QUADRANT (2 co-subtiles x 5
+// k-subtiles = 40 f32/lane; px-split waves would hold 160 accumulators
+// and VGPR-cap occupancy at 1 wave/SIMD). A-frag = dy^T (row = co,
+// k-slice = 8 px), B-frag gathered straight from the padded band.
+// Partials atomic-add into the L2-resident [CO][KPAD] fp32 tile once
+// per block (spread over 2.5k addresses — NOT the one-address-per-block
+// serialization the round-1 notes warn about).
+extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
+    const bf16* __restrict__ x,    // [N][H][W][CI]
+    const bf16* __restrict__ dy,   // [N][HO][WO][CO]
+    float* __restrict__ dw,        // [CO][KPAD] fp32 (pre-zeroed)
+    int Nb, int H, int W, int HO, int WO, int rows_per_block) {
+  __shared__ bf16 xt[8 * XTROW];
+  __shared__ bf16 Dy[PXPAD * CO];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int nrows = Nb * HO;
+  const int mw = (wave >> 1) * 2;   // this wave's co-subtiles: mw, mw+1
+  const int cw = (wave & 1) * 5;    // this wave's k-subtiles: cw .. cw+4
+
+  f32x4 acc[2][5];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int c = 0; c < 5; ++c) acc[m][c] = (f32x4)0.0f;
+
+  for (int rr = 0; rr < rows_per_block; ++rr) {
+    const int row = blockIdx.x * rows_per_block + rr;
+    if (row >= nrows) break;
+    const int n = row / HO;
+    const int ho = row - n * HO;
+    __syncthreads();  // previous iteration's readers are done
+    stage_band(x, xt, n, ho, H, W, tid, 256);
+    {  // dy row: [WO][CO] bf16 contiguous; WO*CO is a multiple of 4
+      const bf16x4* src = reinterpret_cast<const bf16x4*>(
+          dy + (long long)row * WO * CO);
+      bf16x4* d4 = reinterpret_cast<bf16x4*>(Dy);
+      const int nch = (WO * CO) >> 2;
+      const bf16x4 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
+      for (int i = tid; i < (PXPAD * CO) >> 2; i += 256)
+        d4[i] = (i < nch) ? src[i] : z;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int sl = 0; sl < 4; ++sl) {
+      const int pxbase = sl * 32;
+      bf16x8 afrag[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        bf16x8 f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          f[j] = Dy[(pxbase + kgrp * 8 + j) * CO + (mw + m) * 16 + row16];
+        afrag[m] = f;
+      }
+#pragma unroll
+      for (int c = 0; c < 5; ++c) {
+        const int ko = koff_of((cw + c) * 16 + row16);
+        bf16x8 bfr;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfr[j] = xt[ko + (pxbase + kgrp * 8 + j) * (STRIDE * CI)];
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+          acc[m][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[m], bfr, acc[m][c], 0, 0, 0);
+      }
+    }
+  }
+
+  // D row = co (4*(lane>>4)+r), col = k (c*16 + lane&15)
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int c = 0; c < 5; ++c)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = (mw + m) * 16 + kgrp * 4 + r;
+        const int k = (cw + c) * 16 + row16;
+        atomicAdd(&dw[co * KPAD + k], acc[m][c][r]);
+      }
+}
+
+// dw fp32 [CO][KPAD] -> bf16 grad in the conv weight's layout [CO][KTAP]
+extern "C" __global__ __launch_bounds__(256) void stem_wrw_cast_kernel(
+    const float* __restrict__ dw, bf16* __restrict__ out) {
+  const int i = blockIdx.x * 256 + threadIdx.x;
+  if (i < CO * KTAP) {
+    const int co = i / KTAP, k = i - co * KTAP;
+    out[i] = (bf16)dw[co * KPAD + k];
+  }
+}
+
+extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
+                                     int Nb, int H, int W, int HO, int WO,
+                                     hipStream_t stream, int phase_mask) {
+  hipLaunchKernelGGL(stem_conv_fwd_kernel, dim3(Nb * HO), dim3(256), 0,
+                     stream, (const bf16*)x, (const bf16*)w, (bf16*)out,
+                     Nb, H, W, HO, WO, phase_mask);
+}
+
+extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
+                                     float* dw_f32, void* dw_bf16,
+                                     int Nb, int H, int W, int HO, int WO,
+                                     hipStream_t stream) {
+  const int nrows = Nb * HO;
+  const int target_blocks = 768;
+  const int rpb = (nrows + target_blocks - 1) / target_blocks;
+  const int blocks = (nrows + rpb - 1) / rpb;
+  hipMemsetAsync(dw_f32, 0, CO * KPAD * sizeof(float), stream);
+  hipLaunchKernelGGL(stem_conv_wrw_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (const bf16*)x, (const bf16*)dy, dw_f32,
+                     Nb, H, W, HO, WO, rpb);
+  hipLaunchKernelGGL(stem_wrw_cast_kernel,
+                     dim3((CO * KTAP + 255) / 256), dim3(256), 0, stream,
+                     dw_f32, (bf16*)dw_bf16);
+}

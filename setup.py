@@ -22,6 +22,7 @@ SRC = [
     "mi355x_scale/ops/csrc/adam.hip",
     "mi355x_scale/ops/csrc/maxpool.hip",
     "mi355x_scale/ops/csrc/arma_gen.hip",
+    "mi355x_scale/ops/csrc/stemconv.hip",
 ]
 
 ASAN = os.environ.get("MI355X_ASAN") == "1"
