@@ -66,7 +66,7 @@ extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
 extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
                                      float* dw_f32, void* dw_bf16,
                                      int Nb, int H, int W, int HO, int WO,
-                                     hipStream_t stream);
+                                     hipStream_t stream, int phase_mask);
 
 static void _check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
@@ -485,7 +485,7 @@ void stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
 // Weight grad: dw (bf16, weight layout) from x and dy; dw_f32 is the
 // [64][160] fp32 accumulation scratch.
 void stem_conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dw_f32,
-                   torch::Tensor dw) {
+                   torch::Tensor dw, int64_t phase_mask = 7) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(dw_f32.scalar_type() == torch::kFloat32 &&
@@ -497,7 +497,8 @@ void stem_conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dw_f32,
   launch_stem_conv_wrw(x.data_ptr(), dy.data_ptr(),
                        dw_f32.data_ptr<float>(), dw.data_ptr(),
                        Nb, H, W, HO, WO,
-                       at::cuda::getCurrentHIPStream().stream());
+                       at::cuda::getCurrentHIPStream().stream(),
+                       (int)phase_mask);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -506,7 +507,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("out"),
         pybind11::arg("phase_mask") = 7);
   m.def("stem_conv_wrw", &stem_conv_wrw,
-        "MFMA stem conv weight-grad (fp32 accum + bf16 cast)");
+        "MFMA stem conv weight-grad (fp32 accum + bf16 cast)",
+        pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("dw_f32"),
+        pybind11::arg("dw"), pybind11::arg("phase_mask") = 7);
   m.def("normalize_u8_to_bf16", &normalize_u8_to_bf16,
         "fused uint8 NHWC -> normalized bf16 (same memory order)");
   m.def("groupfit_eval", &groupfit_eval,

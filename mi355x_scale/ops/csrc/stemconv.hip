@@ -50,27 +50,61 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-// Zero the whole 8-row band, then copy the 7 valid rows (short4 vector
-// copies when the source rows are 8B-aligned). Callers barrier before
-// the MFMA phase.
+// Stage the 7-row band. ``prime``: zero the whole 8-row tile first (the
+// data region is fully overwritten every call with the same W, so
+// non-prime calls only re-zero row slots whose ih fell outside the
+// image — the lead/tail pads and row 7 stay zero from the prime).
+// Copies are flattened over (row, chunk) and register-batched 5 deep so
+// the global loads pipeline instead of paying one latency each (the v2
+// lesson). Callers barrier before the MFMA phase.
 __device__ __forceinline__ void stage_band(
     const bf16* __restrict__ x, bf16* xt, int n, int ho, int H, int W,
-    int tid, int nthreads) {
-  for (int i = tid; i < 8 * XTROW; i += nthreads) xt[i] = (bf16)0.0f;
-  __syncthreads();
+    int tid, int nthreads, bool prime) {
   const int rowlen = W * CI;
-  const bool vec_ok = (rowlen & 3) == 0;
-  for (int r = 0; r < KH; ++r) {
-    const int ih = ho * STRIDE - PADDING + r;
-    if (ih < 0 || ih >= H) continue;
-    const bf16* src = x + ((long long)n * H + ih) * rowlen;
-    bf16* dst = xt + r * XTROW + XTOFF;
-    if (vec_ok) {
-      const bf16x4* s4 = reinterpret_cast<const bf16x4*>(src);
-      bf16x4* d4 = reinterpret_cast<bf16x4*>(dst);
-      const int nch = rowlen >> 2;
-      for (int i = tid; i < nch; i += nthreads) d4[i] = s4[i];
-    } else {
+  if (prime) {
+    for (int i = tid; i < 8 * XTROW; i += nthreads) xt[i] = (bf16)0.0f;
+  } else {
+    for (int r = 0; r < KH; ++r) {
+      const int ih = ho * STRIDE - PADDING + r;
+      if (ih < 0 || ih >= H)
+        for (int i = tid; i < rowlen; i += nthreads)
+          xt[r * XTROW + XTOFF + i] = (bf16)0.0f;
+    }
+  }
+  __syncthreads();
+  if ((rowlen & 3) == 0) {
+    const int nch = rowlen >> 2;
+    const int total = KH * nch;
+    const bf16x4 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
+    for (int base = 0; base < total; base += nthreads * 5) {
+      bf16x4 v[5];
+      int rr[5], cc[5];
+#pragma unroll
+      for (int u = 0; u < 5; ++u) {
+        const int i = base + tid + u * nthreads;
+        v[u] = z;
+        rr[u] = -1;
+        if (i < total) {
+          const int r = i / nch, c = i - r * nch;
+          const int ih = ho * STRIDE - PADDING + r;
+          rr[u] = r; cc[u] = c;
+          if (ih >= 0 && ih < H)
+            v[u] = reinterpret_cast<const bf16x4*>(
+                x + ((long long)n * H + ih) * rowlen)[c];
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < 5; ++u)
+        if (rr[u] >= 0)
+          reinterpret_cast<bf16x4*>(
+              xt + rr[u] * XTROW + XTOFF)[cc[u]] = v[u];
+    }
+  } else {  // odd row length (test shapes): scalar fallback
+    for (int r = 0; r < KH; ++r) {
+      const int ih = ho * STRIDE - PADDING + r;
+      if (ih < 0 || ih >= H) continue;
+      const bf16* src = x + ((long long)n * H + ih) * rowlen;
+      bf16* dst = xt + r * XTROW + XTOFF;
       for (int i = tid; i < rowlen; i += nthreads) dst[i] = src[i];
     }
   }
@@ -124,7 +158,7 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_fwd_kernel(
         }
       }
     }
-    stage_band(x, xt, n, ho, H, W, tid, 256);
+    stage_band(x, xt, n, ho, H, W, tid, 256, true);
   }
   __syncthreads();
   if (!(phase_mask & 4)) return;
@@ -192,7 +226,8 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
     const bf16* __restrict__ x,    // [N][H][W][CI]
     const bf16* __restrict__ dy,   // [N][HO][WO][CO]
     float* __restrict__ dw,        // [CO][KPAD] fp32 (pre-zeroed)
-    int Nb, int H, int W, int HO, int WO, int rows_per_block) {
+    int Nb, int H, int W, int HO, int WO, int rows_per_block,
+    int phase_mask) {
   __shared__ bf16 xt[8 * XTROW];
   __shared__ bf16 Dy[PXPAD * CO];
   const int tid = threadIdx.x;
@@ -216,17 +251,27 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
     const int n = row / HO;
     const int ho = row - n * HO;
     __syncthreads();  // previous iteration's readers are done
-    stage_band(x, xt, n, ho, H, W, tid, 256);
-    {  // dy row: [WO][CO] bf16 contiguous; WO*CO is a multiple of 4
-      const bf16x4* src = reinterpret_cast<const bf16x4*>(
+    if (phase_mask & 1) stage_band(x, xt, n, ho, H, W, tid, 256, rr == 0);
+    if (phase_mask & 2) {
+      // dy row: [WO][CO] bf16 contiguous, 16B-aligned rows; 16B chunks
+      // register-batched 4 deep (4 x 256 x 8 = PXPAD*CO exactly)
+      const bf16x8* src = reinterpret_cast<const bf16x8*>(
           dy + (long long)row * WO * CO);
-      bf16x4* d4 = reinterpret_cast<bf16x4*>(Dy);
-      const int nch = (WO * CO) >> 2;
-      const bf16x4 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
-      for (int i = tid; i < (PXPAD * CO) >> 2; i += 256)
-        d4[i] = (i < nch) ? src[i] : z;
+      bf16x8* d8 = reinterpret_cast<bf16x8*>(Dy);
+      const int nch = (WO * CO) >> 3;
+      const bf16x8 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f,
+                        (bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
+      bf16x8 v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int i = tid + u * 256;
+        v[u] = (i < nch) ? src[i] : z;
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) d8[tid + u * 256] = v[u];
     }
     __syncthreads();
+    if (!(phase_mask & 4)) continue;
 
 #pragma unroll
     for (int sl = 0; sl < 4; ++sl) {
@@ -289,7 +334,8 @@ extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
 extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
                                      float* dw_f32, void* dw_bf16,
                                      int Nb, int H, int W, int HO, int WO,
-                                     hipStream_t stream) {
+                                     hipStream_t stream,
+                                     int phase_mask = 7) {
   const int nrows = Nb * HO;
   const int target_blocks = 768;
   const int rpb = (nrows + target_blocks - 1) / target_blocks;
@@ -297,7 +343,7 @@ extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
   hipMemsetAsync(dw_f32, 0, CO * KPAD * sizeof(float), stream);
   hipLaunchKernelGGL(stem_conv_wrw_kernel, dim3(blocks), dim3(256), 0,
                      stream, (const bf16*)x, (const bf16*)dy, dw_f32,
-                     Nb, H, W, HO, WO, rpb);
+                     Nb, H, W, HO, WO, rpb, phase_mask);
   hipLaunchKernelGGL(stem_wrw_cast_kernel,
                      dim3((CO * KTAP + 255) / 256), dim3(256), 0, stream,
                      dw_f32, (bf16*)dw_bf16);
