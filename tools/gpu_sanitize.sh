@@ -49,6 +49,25 @@ print("ASAN probe ok:", y.shape, y.dtype)
 EOF
 PROBE_RC=$?
 echo "probe rc=$PROBE_RC" | tee -a "$LOG"
+if [ $PROBE_RC -ne 0 ]; then
+  # gcc-libasan's __cxa_throw interceptor needs the C++ runtime resolved
+  # before torch's lazy dlopen chain — retry with libstdc++ pre-bound
+  STDCXX=$(ldconfig -p | awk '/libstdc\+\+\.so\.6 \(/ {print $NF; exit}')
+  echo "retry with libstdc++ preloaded: $STDCXX" | tee -a "$LOG"
+  LD_PRELOAD="$STDCXX:$ASAN_RT" timeout 240 python - >> "$LOG" 2>&1 <<'EOF'
+import torch
+from mi355x_scale.ops import _C, HAVE_EXT
+assert HAVE_EXT and "_C_asan" in _C.__file__, _C
+x = torch.randint(0, 255, (4, 16, 16, 3), dtype=torch.uint8, device="cuda")
+from mi355x_scale.ops import normalize_images
+y = normalize_images(x)
+torch.cuda.synchronize()
+print("ASAN probe ok (stdc++ preload):", y.shape, y.dtype)
+EOF
+  PROBE_RC=$?
+  echo "probe2 rc=$PROBE_RC" | tee -a "$LOG"
+  [ $PROBE_RC -eq 0 ] && ASAN_RT="$STDCXX:$ASAN_RT"
+fi
 
 if [ $PROBE_RC -eq 0 ]; then
   echo "== device-ASAN kernel suite ($(date -u +%FT%TZ)) ==" | tee -a "$LOG"
