@@ -61,6 +61,7 @@ extern "C" void launch_groupfit_final(
     int T, long long G, int KX, hipStream_t stream);
 
 extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
+                                     void* wp_scratch,
                                      int Nb, int H, int W, int HO, int WO,
                                      hipStream_t stream, int phase_mask);
 extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
@@ -467,7 +468,7 @@ void adam_step_mixed(torch::Tensor master, torch::Tensor gb,
 // x [N,H,W,3] bf16 channels-last storage, w [64,3,7,7] channels_last,
 // out [N,HO,WO,64].
 void stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
-                   int64_t phase_mask = 7) {
+                   torch::Tensor wp_scratch, int64_t phase_mask = 7) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(out.is_cuda() && out.scalar_type() == torch::kBFloat16);
@@ -476,7 +477,10 @@ void stem_conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
   int HO = out.size(1), WO = out.size(2);
   TORCH_CHECK(WO <= 128, "stem kernel handles output rows up to 128 px");
   TORCH_CHECK(w.numel() == 64 * 7 * 7 * 3);
+  TORCH_CHECK(wp_scratch.scalar_type() == torch::kBFloat16 &&
+              wp_scratch.numel() == 64 * 160);
   launch_stem_conv_fwd(x.data_ptr(), w.data_ptr(), out.data_ptr(),
+                       wp_scratch.data_ptr(),
                        Nb, H, W, HO, WO,
                        at::cuda::getCurrentHIPStream().stream(),
                        (int)phase_mask);
@@ -505,7 +509,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stem_conv_fwd", &stem_conv_fwd,
         "MFMA stem conv fwd (7x7 s2, 3->64, NHWC bf16)",
         pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("out"),
-        pybind11::arg("phase_mask") = 7);
+        pybind11::arg("wp_scratch"), pybind11::arg("phase_mask") = 7);
   m.def("stem_conv_wrw", &stem_conv_wrw,
         "MFMA stem conv weight-grad (fp32 accum + bf16 cast)",
         pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("dw_f32"),

@@ -45,9 +45,13 @@ __global__ __launch_bounds__(256) void maxpool_fwd_kernel(
 
   for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
        row < out_rows; row += rstride) {
-    int wo = (int)(row % Wo);
-    int ho = (int)((row / Wo) % Ho);
-    int n = (int)(row / ((long long)Ho * Wo));
+    // 32-bit decomposition: the 64-bit / and % pair is a ~300-cycle
+    // software routine and was a measurable slice of the kernel
+    const int r32 = (int)row;
+    const int wo = r32 % Wo;
+    const int t32 = r32 / Wo;
+    const int ho = t32 % Ho;
+    const int n = t32 / Ho;
     int h0 = 2 * ho - 1, w0 = 2 * wo - 1;
 
     float best[VEC];
@@ -114,9 +118,11 @@ __global__ __launch_bounds__(256) void maxpool_bwd_kernel(
 
   for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
        row < tiles; row += rstride) {
-    int j = (int)(row % Wo);
-    int i = (int)((row / Wo) % Ho);
-    int n = (int)(row / ((long long)Ho * Wo));
+    const int r32 = (int)row;
+    const int j = r32 % Wo;
+    const int t32 = r32 / Wo;
+    const int i = t32 % Ho;
+    const int n = t32 / Ho;
     const long long lvec = (long long)lane * VEC;
 
     // load the (up to) four windows' dy + code vectors

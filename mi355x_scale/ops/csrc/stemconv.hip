@@ -110,6 +110,60 @@ __device__ __forceinline__ void stage_band(
   }
 }
 
+// Ring-band staging for the wrw row loop: consecutive output rows (same
+// n) share 5 of their 7 input rows, so after a prime only the TWO new
+// rows are staged per step. Row slot = ih & 7 (bitwise works for the
+// negative top-edge ihs); invalid rows are zeroed in their slot. The
+// K-padding taps (kh = 7) would alias a live slot, so the wrw fragment
+// builder zero-selects k >= KTAP instead of reading a dedicated row.
+__device__ __forceinline__ void stage_band_ring(
+    const bf16* __restrict__ x, bf16* xt, int n, int ho, int H, int W,
+    int tid, int nthreads, bool prime) {
+  const int rowlen = W * CI;
+  const int ih0 = ho * STRIDE - PADDING;
+  const int rfirst = prime ? 0 : KH - STRIDE;  // prime: all 7; else last 2
+  if ((rowlen & 3) == 0) {
+    const int nch = rowlen >> 2;
+    const int nrows = KH - rfirst;
+    const int total = nrows * nch;
+    const bf16x4 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
+    for (int base = 0; base < total; base += nthreads * 5) {
+      bf16x4 v[5];
+      int ss[5], cc[5];
+#pragma unroll
+      for (int u = 0; u < 5; ++u) {
+        const int i = base + tid + u * nthreads;
+        v[u] = z;
+        ss[u] = -1;
+        if (i < total) {
+          const int r = rfirst + i / nch, c = i - (i / nch) * nch;
+          const int ih = ih0 + r;
+          ss[u] = (ih & 7); cc[u] = c;
+          if (ih >= 0 && ih < H)
+            v[u] = reinterpret_cast<const bf16x4*>(
+                x + ((long long)n * H + ih) * rowlen)[c];
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < 5; ++u)
+        if (ss[u] >= 0)
+          reinterpret_cast<bf16x4*>(
+              xt + ss[u] * XTROW + XTOFF)[cc[u]] = v[u];
+    }
+  } else {
+    for (int r = rfirst; r < KH; ++r) {
+      const int ih = ih0 + r;
+      bf16* dst = xt + (ih & 7) * XTROW + XTOFF;
+      if (ih >= 0 && ih < H) {
+        const bf16* src = x + ((long long)n * H + ih) * rowlen;
+        for (int i = tid; i < rowlen; i += nthreads) dst[i] = src[i];
+      } else {
+        for (int i = tid; i < rowlen; i += nthreads) dst[i] = (bf16)0.0f;
+      }
+    }
+  }
+}
+
 // koff(k): element offset of tap k for pixel 0 in the padded band.
 // tap iw = px*STRIDE - PADDING + kw -> element
 // XTOFF + (px*STRIDE - PADDING + kw)*CI + ci
@@ -121,17 +175,30 @@ __device__ __forceinline__ int koff_of(int k) {
   return kh * XTROW + (XTOFF - PADDING * CI) + r21;
 }
 
+// Pad the weights once per forward: [CO][KTAP] -> [CO][KPAD] (zeros in
+// the MFMA K-padding taps). 20 KB — L1-resident for every fwd block, so
+// the per-block LDS weight-tile refill (40 serialized loads/lane x
+// 23744 blocks) disappears entirely.
+extern "C" __global__ __launch_bounds__(256) void stem_pad_weights_kernel(
+    const bf16* __restrict__ w, bf16* __restrict__ wp) {
+  const int i = blockIdx.x * 256 + threadIdx.x;
+  if (i < CO * KPAD) {
+    const int co = i / KPAD, k = i - co * KPAD;
+    wp[i] = (k < KTAP) ? w[co * KTAP + k] : (bf16)0.0f;
+  }
+}
+
 // ---------------------------------------------------------------- forward
 // block = 256 threads (4 waves) = one output row; wave w computes px
-// subtiles {2w, 2w+1}. LDS = band (12.8 KB) + weight tile (21 KB) ->
-// 4 blocks/CU resident.
+// subtiles {2w, 2w+1}. LDS = band only (13 KB) -> many blocks/CU; the
+// B-fragments read the padded weight buffer straight from global (the
+// whole 20 KB tile lives in L1).
 extern "C" __global__ __launch_bounds__(256) void stem_conv_fwd_kernel(
     const bf16* __restrict__ x,   // [N][H][W][CI]
-    const bf16* __restrict__ w,   // [CO][KH][KW][CI] (channels_last)
+    const bf16* __restrict__ wp,  // [CO][KPAD] padded (stem_pad_weights)
     bf16* __restrict__ out,       // [N][HO][WO][CO]
     int Nb, int H, int W, int HO, int WO, int phase_mask) {
   __shared__ bf16 xt[8 * XTROW];
-  __shared__ bf16 Wl[CO * KLDS];
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -139,25 +206,6 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_fwd_kernel(
   const int ho = blockIdx.x - n * HO;
 
   if (phase_mask & 1) {
-    // weight tile: 40 els/thread, loads batched 8 deep
-    for (int base = tid * 8; base < CO * KPAD; base += 256 * 8) {
-      bf16 v[8];
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int idx = base + u;
-        const int co = idx / KPAD, k = idx - co * KPAD;
-        v[u] = (idx < CO * KPAD && k < KTAP) ? w[co * KTAP + k]
-                                             : (bf16)0.0f;
-      }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int idx = base + u;
-        if (idx < CO * KPAD) {
-          const int co = idx / KPAD, k = idx - co * KPAD;
-          Wl[co * KLDS + k] = v[u];
-        }
-      }
-    }
     stage_band(x, xt, n, ho, H, W, tid, 256, true);
   }
   __syncthreads();
@@ -179,7 +227,7 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_fwd_kernel(
 #pragma unroll
     for (int c = 0; c < 4; ++c)
       bfrag[c] = *reinterpret_cast<const bf16x8*>(
-          Wl + (c * 16 + row16) * KLDS + kk * 32 + kgrp * 8);
+          wp + (c * 16 + row16) * KPAD + kk * 32 + kgrp * 8);
     bf16x8 f0, f1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -250,8 +298,13 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
     if (row >= nrows) break;
     const int n = row / HO;
     const int ho = row - n * HO;
+    const bool prime = (rr == 0) || (ho == 0);
     __syncthreads();  // previous iteration's readers are done
-    if (phase_mask & 1) stage_band(x, xt, n, ho, H, W, tid, 256, rr == 0);
+    if (prime)  // lead/tail pads must be zero before the first data rows
+      for (int i = tid; i < 8 * XTROW; i += 256) xt[i] = (bf16)0.0f;
+    __syncthreads();
+    if (phase_mask & 1)
+      stage_band_ring(x, xt, n, ho, H, W, tid, 256, prime);
     if (phase_mask & 2) {
       // dy row: [WO][CO] bf16 contiguous, 16B-aligned rows; 16B chunks
       // register-batched 4 deep (4 x 256 x 8 = PXPAD*CO exactly)
@@ -287,11 +340,18 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
       }
 #pragma unroll
       for (int c = 0; c < 5; ++c) {
-        const int ko = koff_of((cw + c) * 16 + row16);
+        const int k = (cw + c) * 16 + row16;
+        const int kh = k / (KW * CI);
+        const int r21 = k - kh * (KW * CI);
+        // ring slot for this tap's input row; pad taps zero-selected
+        const int ko = ((ho * STRIDE - PADDING + kh) & 7) * XTROW +
+                       (XTOFF - PADDING * CI) + r21;
+        const bool kz = k >= KTAP;
         bf16x8 bfr;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          bfr[j] = xt[ko + (pxbase + kgrp * 8 + j) * (STRIDE * CI)];
+          bfr[j] = kz ? (bf16)0.0f
+                      : xt[ko + (pxbase + kgrp * 8 + j) * (STRIDE * CI)];
 #pragma unroll
         for (int m = 0; m < 2; ++m)
           acc[m][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -324,11 +384,15 @@ extern "C" __global__ __launch_bounds__(256) void stem_wrw_cast_kernel(
 }
 
 extern "C" void launch_stem_conv_fwd(const void* x, const void* w, void* out,
+                                     void* wp_scratch,
                                      int Nb, int H, int W, int HO, int WO,
                                      hipStream_t stream, int phase_mask) {
+  hipLaunchKernelGGL(stem_pad_weights_kernel,
+                     dim3((CO * KPAD + 255) / 256), dim3(256), 0, stream,
+                     (const bf16*)w, (bf16*)wp_scratch);
   hipLaunchKernelGGL(stem_conv_fwd_kernel, dim3(Nb * HO), dim3(256), 0,
-                     stream, (const bf16*)x, (const bf16*)w, (bf16*)out,
-                     Nb, H, W, HO, WO, phase_mask);
+                     stream, (const bf16*)x, (const bf16*)wp_scratch,
+                     (bf16*)out, Nb, H, W, HO, WO, phase_mask);
 }
 
 extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,

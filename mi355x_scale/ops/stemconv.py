@@ -34,8 +34,9 @@ class _StemConvFn(torch.autograd.Function):
                           device=x.device,
                           memory_format=torch.channels_last)
         # kernels address raw NHWC storage: hand the physical views over
+        wp = torch.empty(64 * 160, dtype=torch.bfloat16, device=x.device)
         _C.stem_conv_fwd(x.permute(0, 2, 3, 1), weight.permute(0, 2, 3, 1),
-                         out.permute(0, 2, 3, 1))
+                         out.permute(0, 2, 3, 1), wp)
         ctx.save_for_backward(x, weight)
         return out
 
