@@ -17,12 +17,16 @@ from typing import List, Optional, Type, Union
 import torch
 import torch.nn as nn
 
+from ..ops.conv3x3 import Conv3x3
 from ..ops.fused_bn import FusedBNReLU2d
 from ..ops.maxpool import MaxPool3x3s2
 from ..ops.stemconv import StemConv2d
 
 
 def _conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    if stride == 1 and cin == cout:
+        # MFMA weight-grad fast path (ops/csrc/conv3x3wrw.hip)
+        return Conv3x3(cin, cout)
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 

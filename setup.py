@@ -23,6 +23,7 @@ SRC = [
     "mi355x_scale/ops/csrc/maxpool.hip",
     "mi355x_scale/ops/csrc/arma_gen.hip",
     "mi355x_scale/ops/csrc/stemconv.hip",
+    "mi355x_scale/ops/csrc/conv3x3wrw.hip",
 ]
 
 ASAN = os.environ.get("MI355X_ASAN") == "1"
