@@ -72,6 +72,18 @@ def main():
         mse_gpu = float(np.mean(
             (out_gpu["Demand"] - out_gpu["Demand_Fitted"]) ** 2))
         print(f"GPU batched path: panel MSE {mse_gpu:.1f}")
+
+    # 5. restartable persisted output (the reference's Delta write,
+    # group_apply/02_...py:544-552): per-group-shard Parquet + atomic
+    # rename — a killed job resumes without refitting finished shards
+    import tempfile
+    from mi355x_scale.forecast import (read_forecast_shards,
+                                       run_fine_grained_forecast_sharded)
+    with tempfile.TemporaryDirectory() as d:
+        run_fine_grained_forecast_sharded(df, d, num_shards=4,
+                                          max_evals=4)
+        persisted = read_forecast_shards(d)
+        print(f"sharded output: {len(persisted)} rows across 4 shards")
     return out
 
 

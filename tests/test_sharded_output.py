@@ -64,6 +64,24 @@ def test_resume_skips_finished_shards_byte_identical(tmp_path, demand):
     assert len(frame) == len(demand)
 
 
+@pytest.mark.gpu
+def test_sharded_gpu_engine_resume(tmp_path, demand):
+    """The GPU (batched-kernel) engine through the sharded writer:
+    complete coverage, resume-by-shard, deterministic bytes."""
+    out = str(tmp_path / "fcg")
+    paths = run_fine_grained_forecast_sharded(
+        demand, out, num_shards=3, engine="gpu", horizon=12)
+    frame = read_forecast_shards(out)
+    assert len(frame) == len(demand)
+    assert frame["Demand_Fitted"].notna().all()
+    blobs = {p: open(p, "rb").read() for p in paths}
+    os.unlink(paths[1])
+    run_fine_grained_forecast_sharded(
+        demand, out, num_shards=3, engine="gpu", horizon=12)
+    for p in paths:
+        assert open(p, "rb").read() == blobs[p]
+
+
 def test_rank_sharding_partitions_work(tmp_path, demand):
     out = str(tmp_path / "fc")
     _run(demand, out, cur_rank=0, world_size=2)
