@@ -35,7 +35,8 @@ def main():
             .to(torch.bfloat16).to(memory_format=torch.channels_last)
         wt = (torch.randn(C, C, 3, 3, generator=g) * 0.05).cuda() \
             .to(torch.bfloat16).to(memory_format=torch.channels_last)
-        scratch = torch.empty(C * 9 * C, dtype=torch.float32,
+        nch = _C.conv3x3_wrw_nchunks(212, HW, C, 0)
+        scratch = torch.empty(nch * C * 9 * C, dtype=torch.float32,
                               device="cuda")
         dw = torch.empty_like(wt)
 

@@ -10,12 +10,16 @@ pattern), so the add disappears from the captured step.
 
 STATUS (round 2, measured on MI355X — benchmarks/bench_conv3x3_wrw.py):
 numerics verified against the fp32 reference on every block-conv shape,
-but the kernel is SLOWER than MIOpen igemm wrw (0.56x at C=64 56²,
-down to 0.08x at C=512 7²): the 32x32 dW tile duplicates the x/dy
-staging across (C/32)² tile combos, which overwhelms the k-loop for
-large C. Default-OFF (``MI355X_CONV_WRW=1`` opts in); the fix —
-64x64+ block tiles with full GEMM-style staging discipline — is the
-top round-3 lead (NOTES_NEXT.md).
+but two blocking iterations both lost to MIOpen igemm wrw (v1 32x32
+tiles: 0.56x at C=64; v2 64x64 tiles / 8-wave blocks / non-atomic chunk
+partials: 0.35x at C=64 down to 0.10x at C=512). Ablations localize the
+cost to the thin per-row phase structure: one staging latency + three
+barriers per 56-px output row, at 1 block/CU (211 VGPRs -> 2
+waves/SIMD), cannot fill the machine, and the K-padding waste at
+W<=14 is fatal for layer3/4. Default-OFF (``MI355X_CONV_WRW=1`` opts
+in); the round-3 path is software-pipelined multi-row K-phases (stage
+row r+1 into registers during row r's MFMAs) plus a VGPR diet — see
+NOTES_NEXT.md.
 
 ``Conv3x3`` subclasses ``nn.Conv2d`` (state-dict compatible, same
 init). When enabled, the HIP path takes CUDA bf16 channels-last with
@@ -50,7 +54,8 @@ class _Conv3x3Fn(torch.autograd.Function):
                 dy, x, weight, None, [1, 1], [1, 1], [1, 1], False,
                 [0, 0], 1, [True, False, False])[0]
         C = weight.shape[0]
-        scratch = torch.empty(C * 9 * C, dtype=torch.float32,
+        nch = _C.conv3x3_wrw_nchunks(x.shape[0], x.shape[2], C, 0)
+        scratch = torch.empty(nch * C * 9 * C, dtype=torch.float32,
                               device=x.device)
         g = weight.grad
         fuse = (torch.cuda.is_current_stream_capturing()
@@ -76,7 +81,7 @@ class Conv3x3(nn.Conv2d):
         return (HAVE_EXT and x.is_cuda
                 and os.environ.get("MI355X_CONV_WRW", "0") == "1"
                 and self.in_channels == C
-                and C % 32 == 0 and C <= 2048
+                and C % 64 == 0 and C <= 2048
                 and 9 <= x.shape[-1] <= 64
                 and x.dtype == torch.bfloat16
                 and self.weight.dtype == torch.bfloat16

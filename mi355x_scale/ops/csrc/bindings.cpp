@@ -69,10 +69,11 @@ extern "C" void launch_stem_conv_wrw(const void* x, const void* dy,
                                      int Nb, int H, int W, int HO, int WO,
                                      hipStream_t stream, int phase_mask);
 extern "C" void launch_conv3x3_wrw(const void* x, const void* dy,
-                                   float* dwf, void* dw_bf16,
+                                   float* part, void* dw_bf16,
                                    int Nb, int H, int W, int C,
-                                   hipStream_t stream, int phase_mask,
-                                   int chunks_override);
+                                   int chunks, hipStream_t stream,
+                                   int phase_mask);
+extern "C" int conv3x3_wrw_chunks(int Nb, int H, int C, int chunks_req);
 
 static void _check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
@@ -510,9 +511,10 @@ void stem_conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dw_f32,
                        (int)phase_mask);
 }
 
-// 3x3/s1/p1 CxC weight grad: dwf fp32 [C][9][C] scratch, dw bf16 out
-// in channels_last weight layout (may be the flat grad view).
-void conv3x3_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dwf,
+// 3x3/s1/p1 CxC weight grad: part fp32 [chunks][C][9][C] scratch
+// (size via conv3x3_wrw_nchunks), dw bf16 out in channels_last weight
+// layout (may be the flat grad view).
+void conv3x3_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor part,
                  torch::Tensor dw, int64_t phase_mask = 7,
                  int64_t chunks = 0) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
@@ -520,24 +522,33 @@ void conv3x3_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor dwf,
   int Nb = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   TORCH_CHECK(dy.size(1) == H && dy.size(2) == W && dy.size(3) == C,
               "same-shape 3x3 s1 conv only");
-  TORCH_CHECK(C % 32 == 0 && C <= 2048, "C must be a multiple of 32");
+  TORCH_CHECK(C % 64 == 0 && C <= 2048, "C must be a multiple of 64");
   TORCH_CHECK(W >= 3 && W <= 64, "row length outside kernel bounds");
-  TORCH_CHECK(dwf.scalar_type() == torch::kFloat32 &&
-              dwf.numel() == (long long)C * 9 * C);
+  const int nch = conv3x3_wrw_chunks(Nb, H, C, (int)chunks);
+  TORCH_CHECK(part.scalar_type() == torch::kFloat32 &&
+              part.numel() >= (long long)nch * C * 9 * C,
+              "partial buffer too small for ", nch, " chunks");
   TORCH_CHECK(dw.scalar_type() == torch::kBFloat16 &&
               dw.numel() == (long long)C * 9 * C);
-  launch_conv3x3_wrw(x.data_ptr(), dy.data_ptr(), dwf.data_ptr<float>(),
-                     dw.data_ptr(), Nb, H, W, C,
+  launch_conv3x3_wrw(x.data_ptr(), dy.data_ptr(), part.data_ptr<float>(),
+                     dw.data_ptr(), Nb, H, W, C, nch,
                      at::cuda::getCurrentHIPStream().stream(),
-                     (int)phase_mask, (int)chunks);
+                     (int)phase_mask);
+}
+
+int64_t conv3x3_wrw_nchunks(int64_t Nb, int64_t H, int64_t C,
+                            int64_t chunks) {
+  return conv3x3_wrw_chunks((int)Nb, (int)H, (int)C, (int)chunks);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_wrw", &conv3x3_wrw,
         "MFMA 3x3/s1 CxC weight grad (fp32 accum, bf16 cast to view)",
-        pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("dwf"),
+        pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("part"),
         pybind11::arg("dw"), pybind11::arg("phase_mask") = 7,
         pybind11::arg("chunks") = 0);
+  m.def("conv3x3_wrw_nchunks", &conv3x3_wrw_nchunks,
+        "partial-buffer chunk count the wrw launcher will use");
   m.def("stem_conv_fwd", &stem_conv_fwd,
         "MFMA stem conv fwd (7x7 s2, 3->64, NHWC bf16)",
         pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("out"),
