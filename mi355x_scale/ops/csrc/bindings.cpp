@@ -188,6 +188,20 @@ extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
                                      const float* invstd, float* partial,
                                      void* dym, int nblocks, long long rows,
                                      int C, int relu, hipStream_t stream);
+extern "C" void launch_bn_bwd_reduce_rm(const void* dz, const void* x,
+                                        const float* mean,
+                                        const float* invstd,
+                                        const float* weight,
+                                        const float* bias, float* partial,
+                                        void* dym, int nblocks,
+                                        long long rows, int C,
+                                        hipStream_t stream);
+extern "C" void launch_bn_bwd_apply_rm(const void* dz, const void* x,
+                                       const float* mean,
+                                       const float* invstd,
+                                       const float* bias, const float* k,
+                                       void* dx, long long rows, int C,
+                                       hipStream_t stream);
 extern "C" void launch_bn_bwd_apply_dym(const void* dym, const void* x,
                                         const float* mean,
                                         const float* invstd, const float* k,
@@ -294,6 +308,47 @@ torch::Tensor bn_bwd_reduce(torch::Tensor dz, torch::Tensor y,
                        relu ? 1 : 0,
                        at::cuda::getCurrentHIPStream().stream());
   return partial;
+}
+
+// Recompute-mask variants (non-residual RELU backward): mask from
+// (w*xhat + b) > 0 — no y read, y not saved for backward.
+torch::Tensor bn_bwd_reduce_rm(torch::Tensor dz, torch::Tensor x,
+                               torch::Tensor mean, torch::Tensor invstd,
+                               torch::Tensor weight, torch::Tensor bias,
+                               torch::Tensor dym, int64_t C) {
+  _check_bn_act(dz, "dz"); _check_bn_act(x, "x");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(weight, "weight"); _check_f32(bias, "bias");
+  void* dym_p = nullptr;
+  if (dym.defined() && dym.numel()) {
+    _check_bn_act(dym, "dym");
+    TORCH_CHECK(dym.numel() == x.numel(), "dym shape mismatch");
+    dym_p = dym.data_ptr();
+  }
+  long long rows = _bn_rows(x, C);
+  int nb = bn_reduce_blocks(rows, (int)C);
+  auto partial = torch::empty(
+      {(int64_t)nb * 2 * C},
+      torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+  launch_bn_bwd_reduce_rm(dz.data_ptr(), x.data_ptr(),
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          weight.data_ptr<float>(), bias.data_ptr<float>(),
+                          partial.data_ptr<float>(), dym_p, nb, rows,
+                          (int)C, at::cuda::getCurrentHIPStream().stream());
+  return partial;
+}
+
+void bn_bwd_apply_rm(torch::Tensor dz, torch::Tensor x, torch::Tensor mean,
+                     torch::Tensor invstd, torch::Tensor bias,
+                     torch::Tensor k, torch::Tensor dx, int64_t C) {
+  _check_bn_act(dz, "dz"); _check_bn_act(x, "x"); _check_bn_act(dx, "dx");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(bias, "bias"); _check_f32(k, "k");
+  launch_bn_bwd_apply_rm(dz.data_ptr(), x.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         bias.data_ptr<float>(), k.data_ptr<float>(),
+                         dx.data_ptr(), _bn_rows(x, C), (int)C,
+                         at::cuda::getCurrentHIPStream().stream());
 }
 
 void bn_bwd_apply_dym(torch::Tensor dym, torch::Tensor x,
@@ -571,6 +626,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_finalize", &bn_fwd_finalize);
   m.def("bn_fwd_apply", &bn_fwd_apply);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_reduce_rm", &bn_bwd_reduce_rm);
+  m.def("bn_bwd_apply_rm", &bn_bwd_apply_rm);
   m.def("bn_bwd_finalize", &bn_bwd_finalize);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("bn_bwd_apply_dym", &bn_bwd_apply_dym);

@@ -388,6 +388,124 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
   }
 }
 
+// ---------------------------------------------------- recompute-mask bwd
+// Non-residual RELU backward: the stored y is read ONLY for the relu
+// mask, which is recomputable as (w*xhat + b) > 0 from the x already in
+// registers — one full activation-map read drops out of BOTH backward
+// passes (and y need not be saved for backward at all). The fp32
+// recompute is if anything closer to the fp32 reference than masking on
+// the rounded bf16 y. Residual layers keep the y path (the mask there
+// depends on the residual input).
+template <bool WRITE_DY>
+__global__ __launch_bounds__(256) void bn_bwd_reduce_rm_kernel(
+    const bf16* __restrict__ dz, const bf16* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ weight, const float* __restrict__ bias,
+    float* __restrict__ partial, bf16* __restrict__ dym,
+    long long rows, int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float m[VEC], is[VEC], wv[VEC], bv[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    m[j] = mean[c];
+    is[j] = invstd[c];
+    wv[j] = weight[c];
+    bv[j] = bias[c];
+  }
+
+  float sdy[VEC] = {0.f}, sdyx[VEC] = {0.f};
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec g = load8(dz + off);
+    BVec xv = load8(x + off);
+    BVec dyv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
+      float dy = __bfloat162float(g.h[j]);
+      if (wv[j] * xhat + bv[j] <= 0.f) dy = 0.f;
+      sdy[j] += dy;
+      sdyx[j] += dy * xhat;
+      if (WRITE_DY) dyv.h[j] = __float2bfloat16(dy);
+    }
+    if (WRITE_DY) store8(dym + off, dyv);
+  }
+
+  __shared__ float lds[256 * 2 * VEC];
+  float* mys = &lds[(rsub * lanes + lane) * 2 * VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    mys[j] = sdy[j];
+    mys[VEC + j] = sdyx[j];
+  }
+  __syncthreads();
+  for (int step = rows_per_iter >> 1; step > 0; step >>= 1) {
+    if (rsub < step) {
+      const float* other = &lds[((rsub + step) * lanes + lane) * 2 * VEC];
+#pragma unroll
+      for (int j = 0; j < 2 * VEC; ++j) mys[j] += other[j];
+    }
+    __syncthreads();
+  }
+  if (rsub == 0) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      partial[(long long)(lane * VEC + j) * gridDim.x + blockIdx.x] = mys[j];
+      partial[(long long)(C + lane * VEC + j) * gridDim.x + blockIdx.x] =
+          mys[VEC + j];
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void bn_bwd_apply_rm_kernel(
+    const bf16* __restrict__ dz, const bf16* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ bias, const float* __restrict__ k,
+    bf16* __restrict__ dx, long long rows, int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float m[VEC], is[VEC], wis[VEC], k1[VEC], k2[VEC], bv[VEC], wv[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    int c = lane * VEC + j;
+    m[j] = mean[c];
+    is[j] = invstd[c];
+    wis[j] = k[c];          // w * invstd
+    k1[j] = k[C + c];
+    k2[j] = k[2 * C + c];
+    bv[j] = bias[c];
+    wv[j] = wis[j] / is[j];  // w, recovered once per channel
+  }
+
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < rows; row += rstride) {
+    const long long off = row * C + (long long)lane * VEC;
+    BVec g = load8(dz + off);
+    BVec xv = load8(x + off);
+    BVec odx;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float xhat = (__bfloat162float(xv.h[j]) - m[j]) * is[j];
+      float dy = __bfloat162float(g.h[j]);
+      if (wv[j] * xhat + bv[j] <= 0.f) dy = 0.f;
+      float v = wis[j] * (dy - k1[j] - xhat * k2[j]);
+      odx.h[j] = __float2bfloat16(v);
+    }
+    store8(dx + off, odx);
+  }
+}
+
 // ------------------------------------------------------------------ launchers
 
 static int pick_grid(long long rows, int C) {
@@ -461,6 +579,37 @@ extern "C" void launch_bn_bwd_reduce(const void* dz, const void* y,
   else if (dym) REDUCE(false, true);
   else REDUCE(false, false);
 #undef REDUCE
+}
+
+extern "C" void launch_bn_bwd_reduce_rm(const void* dz, const void* x,
+                                        const float* mean,
+                                        const float* invstd,
+                                        const float* weight,
+                                        const float* bias, float* partial,
+                                        void* dym, int nblocks,
+                                        long long rows, int C,
+                                        hipStream_t stream) {
+  dim3 grid(nblocks), block(256);
+  if (dym)
+    hipLaunchKernelGGL((bn_bwd_reduce_rm_kernel<true>), grid, block, 0,
+                       stream, (const bf16*)dz, (const bf16*)x, mean,
+                       invstd, weight, bias, partial, (bf16*)dym, rows, C);
+  else
+    hipLaunchKernelGGL((bn_bwd_reduce_rm_kernel<false>), grid, block, 0,
+                       stream, (const bf16*)dz, (const bf16*)x, mean,
+                       invstd, weight, bias, partial, nullptr, rows, C);
+}
+
+extern "C" void launch_bn_bwd_apply_rm(const void* dz, const void* x,
+                                       const float* mean,
+                                       const float* invstd,
+                                       const float* bias, const float* k,
+                                       void* dx, long long rows, int C,
+                                       hipStream_t stream) {
+  hipLaunchKernelGGL(bn_bwd_apply_rm_kernel, dim3(pick_grid(rows, C)),
+                     dim3(256), 0, stream, (const bf16*)dz,
+                     (const bf16*)x, mean, invstd, bias, k, (bf16*)dx,
+                     rows, C);
 }
 
 extern "C" void launch_bn_bwd_apply_dym(const void* dym, const void* x,

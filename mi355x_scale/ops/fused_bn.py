@@ -52,7 +52,11 @@ class _FusedBNFn(torch.autograd.Function):
         y = torch.empty_like(x)
         res = residual if residual is not None else x.new_empty(0)
         _C.bn_fwd_apply(x, res, y, mean, invstd, weight, bias, C, relu)
-        ctx.save_for_backward(x, y, mean, invstd, weight, bias)
+        # Non-residual backward recomputes the relu mask from
+        # (w*xhat + b) > 0, so y need not be kept alive; the saved slot
+        # holds x again (free — same tensor).
+        ctx.save_for_backward(x, y if residual is not None else x,
+                              mean, invstd, weight, bias)
         ctx.relu = relu
         ctx.has_res = residual is not None
         return y
@@ -105,12 +109,21 @@ class _FusedBNFn(torch.autograd.Function):
             return (dx, dym, None if fuse_acc else dweight,
                     None if fuse_acc else dbias, None, None, None, None,
                     None)
-        partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd, x.new_empty(0),
-                                   C, ctx.relu)
-        _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
-                           rows, C, fuse_acc)
-        _C.bn_bwd_apply(dz, y, x, mean, invstd, k, dx, x.new_empty(0), C,
-                        ctx.relu)
+        if ctx.relu:
+            # recompute-mask path: one full activation-map read (the
+            # stored y) drops out of BOTH backward passes
+            partial = _C.bn_bwd_reduce_rm(dz, x, mean, invstd, weight,
+                                          bias, x.new_empty(0), C)
+            _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
+                               rows, C, fuse_acc)
+            _C.bn_bwd_apply_rm(dz, x, mean, invstd, bias, k, dx, C)
+        else:
+            partial = _C.bn_bwd_reduce(dz, y, x, mean, invstd,
+                                       x.new_empty(0), C, False)
+            _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
+                               rows, C, fuse_acc)
+            _C.bn_bwd_apply(dz, y, x, mean, invstd, k, dx, x.new_empty(0),
+                            C, False)
         return (dx, None, None if fuse_acc else dweight,
                 None if fuse_acc else dbias, None, None, None, None, None)
 
