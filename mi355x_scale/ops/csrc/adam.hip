@@ -101,9 +101,56 @@ __global__ __launch_bounds__(256) void adam_step_mixed_kernel(
   const float step_size = lr / bc1;
   const long long stride = (long long)gridDim.x * blockDim.x * 4;
 
+  typedef __attribute__((ext_vector_type(4))) short short4v;
   for (long long base = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
        base < n; base += stride) {
     const int lanes = (int)min((long long)4, n - base);
+    const bool all_b = base + 4 <= nb;
+    // fp32-segment float4 loads need (base - nb) 16 B aligned
+    const bool all_f = base >= nb && (nb & 3) == 0;
+    if (lanes == 4 && (all_b || all_f)) {
+      // vectorized fast path (the scalar loop left this kernel at ~2.2x
+      // the HBM roofline): float4 master/m/v, 8 B bf16 grad/param packs
+      float4 pv = *reinterpret_cast<float4*>(master + base);
+      float4 mv = *reinterpret_cast<float4*>(m + base);
+      float4 vv = *reinterpret_cast<float4*>(v + base);
+      float gj[4];
+      if (all_b) {
+        const short4v gs = *reinterpret_cast<const short4v*>(
+            reinterpret_cast<const short*>(gb) + base);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          __hip_bfloat16 h;
+          *reinterpret_cast<short*>(&h) = gs[j];
+          gj[j] = __bfloat162float(h);
+        }
+      } else {
+        const float4 gv = *reinterpret_cast<const float4*>(gf + base - nb);
+        gj[0] = gv.x; gj[1] = gv.y; gj[2] = gv.z; gj[3] = gv.w;
+      }
+      float* pp = reinterpret_cast<float*>(&pv);
+      float* mp = reinterpret_cast<float*>(&mv);
+      float* vp = reinterpret_cast<float*>(&vv);
+      short4v pbv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float g = gj[j] + weight_decay * pp[j];
+        mp[j] = beta1 * mp[j] + (1.f - beta1) * g;
+        vp[j] = beta2 * vp[j] + (1.f - beta2) * g * g;
+        pp[j] -= step_size * mp[j] / (sqrtf(vp[j] / bc2) + eps);
+        if (all_b) {
+          __hip_bfloat16 h = __float2bfloat16(pp[j]);
+          pbv[j] = *reinterpret_cast<short*>(&h);
+        }
+      }
+      *reinterpret_cast<float4*>(master + base) = pv;
+      *reinterpret_cast<float4*>(m + base) = mv;
+      *reinterpret_cast<float4*>(v + base) = vv;
+      if (all_b)
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(pb) + base) = pbv;
+      continue;
+    }
     for (int j = 0; j < lanes; ++j) {
       const long long i = base + j;
       float g = i < nb ? __bfloat162float(gb[i]) : gf[i - nb];
