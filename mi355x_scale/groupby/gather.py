@@ -78,7 +78,10 @@ def fast_factorize(col: pd.Series) -> Tuple[np.ndarray, np.ndarray]:
         if arr.dtype == object:
             try:  # one object->arrow pass; still 4x cheaper than pandas
                 pa_arr = pa.array(col, from_pandas=True)
-            except (pa.ArrowInvalid, pa.ArrowTypeError):
+            except (pa.ArrowInvalid, pa.ArrowTypeError,
+                    UnicodeEncodeError):
+                # UnicodeEncodeError: lone surrogates are valid Python
+                # str but not UTF-8 — pandas' object hashing handles them
                 pa_arr = None
             if pa_arr is not None:
                 res = _factorize_arrow_str(col, pa_arr)
