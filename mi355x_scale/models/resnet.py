@@ -117,7 +117,10 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.maxpool(self.bn1(self.conv1(x)))  # bn1 fuses the relu
+        # stem: bn1's relu+affine fold INTO the pool window reads (the
+        # normalized map is never materialized — ops/fused_bn.py
+        # forward_pooled); composed ops off the GPU bf16 path
+        x = self.bn1.forward_pooled(self.conv1(x))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

@@ -433,6 +433,13 @@ extern "C" void launch_maxpool_fwd(const void* x, void* y,
                                    unsigned char* code, int N, int H, int W,
                                    int Ho, int Wo, int C,
                                    hipStream_t stream);
+extern "C" void launch_bn_maxpool_fwd(const void* x, const float* mean,
+                                      const float* invstd,
+                                      const float* weight,
+                                      const float* bias, void* y,
+                                      void* code, int N, int H, int W,
+                                      int Ho, int Wo, int C,
+                                      hipStream_t stream);
 extern "C" void launch_maxpool_bwd(const void* dy,
                                    const unsigned char* code, void* dx,
                                    int N, int H, int W, int Ho, int Wo,
@@ -452,6 +459,29 @@ void maxpool3x3s2_fwd(torch::Tensor x, torch::Tensor y,
   launch_maxpool_fwd(x.data_ptr(), y.data_ptr(), code.data_ptr<uint8_t>(),
                      N, H, W, Ho, Wo, C,
                      at::cuda::getCurrentHIPStream().stream());
+}
+
+// Fused stem forward: relu(bn(x)) applied inline while pooling — the
+// normalized activation map is never materialized (see maxpool.hip).
+void bn_maxpool3x3s2_fwd(torch::Tensor x, torch::Tensor mean,
+                         torch::Tensor invstd, torch::Tensor weight,
+                         torch::Tensor bias, torch::Tensor y,
+                         torch::Tensor code) {
+  _check_bn_act(x, "x"); _check_bn_act(y, "y");
+  _check_f32(mean, "mean"); _check_f32(invstd, "invstd");
+  _check_f32(weight, "weight"); _check_f32(bias, "bias");
+  TORCH_CHECK(code.scalar_type() == torch::kUInt8 && code.is_cuda());
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Ho = y.size(2), Wo = y.size(3);
+  TORCH_CHECK(C % 8 == 0 && 256 % (C / 8) == 0, "bad C");
+  TORCH_CHECK(Ho == (H + 1) / 2 && Wo == (W + 1) / 2, "3x3 s2 p1 shape");
+  TORCH_CHECK(code.numel() == y.numel());
+  launch_bn_maxpool_fwd(x.data_ptr(), mean.data_ptr<float>(),
+                        invstd.data_ptr<float>(),
+                        weight.data_ptr<float>(), bias.data_ptr<float>(),
+                        y.data_ptr(), code.data_ptr<uint8_t>(),
+                        N, H, W, Ho, Wo, C,
+                        at::cuda::getCurrentHIPStream().stream());
 }
 
 void maxpool3x3s2_bwd(torch::Tensor dy, torch::Tensor code,
@@ -634,6 +664,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step,
         "fused flat Adam step (one kernel over p/g/m/v)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd);
+  m.def("bn_maxpool3x3s2_fwd", &bn_maxpool3x3s2_fwd);
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd);
   m.def("adam_step_mixed", &adam_step_mixed,
         "fused flat Adam with bf16 params/grads + fp32 master");

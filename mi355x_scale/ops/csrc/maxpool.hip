@@ -95,6 +95,84 @@ __global__ __launch_bounds__(256) void maxpool_fwd_kernel(
   }
 }
 
+// Fused stem forward: BatchNorm affine + ReLU applied INLINE to the
+// conv output while pooling — the 340 MB normalized activation map
+// (bs 212) is never materialized. Legal because the recompute-mask BN
+// backward (fused_bn.hip) no longer needs the stored y: the map's only
+// consumer was this pool. Window taps re-read x up to 2.25x, but those
+// hits land in L1/L2; the eliminated full write+read of y is HBM.
+__global__ __launch_bounds__(256) void bn_maxpool_fwd_kernel(
+    const bf16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ weight,
+    const float* __restrict__ bias, bf16* __restrict__ y,
+    unsigned char* __restrict__ code, int N, int H, int W, int Ho, int Wo,
+    int C) {
+  const int lanes = C / VEC;
+  const int lane = threadIdx.x % lanes;
+  const int rsub = threadIdx.x / lanes;
+  const int rows_per_iter = blockDim.x / lanes;
+  const long long out_rows = (long long)N * Ho * Wo;
+  const long long rstride = (long long)gridDim.x * rows_per_iter;
+
+  float sc[VEC], sh[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    const int c = lane * VEC + j;
+    sc[j] = weight[c] * invstd[c];            // y = sc*x + sh, then relu
+    sh[j] = bias[c] - mean[c] * sc[j];
+  }
+
+  for (long long row = (long long)blockIdx.x * rows_per_iter + rsub;
+       row < out_rows; row += rstride) {
+    const int r32 = (int)row;
+    const int wo = r32 % Wo;
+    const int t32 = r32 / Wo;
+    const int ho = t32 % Ho;
+    const int n = t32 / Ho;
+    int h0 = 2 * ho - 1, w0 = 2 * wo - 1;
+
+    float best[VEC];
+    unsigned char bcode[VEC];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      best[j] = -1e30f;
+      bcode[j] = 0;
+    }
+    for (int ih = 0; ih < 3; ++ih) {
+      int h = h0 + ih;
+      if (h < 0 || h >= H) continue;
+      for (int iw = 0; iw < 3; ++iw) {
+        int w = w0 + iw;
+        if (w < 0 || w >= W) continue;
+        const long long off =
+            (((long long)n * H + h) * W + w) * C + (long long)lane * VEC;
+        BVec v;
+        v.u = *reinterpret_cast<const uint4*>(x + off);
+        unsigned char pc = (unsigned char)(ih * 3 + iw);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          float f = sc[j] * __bfloat162float(v.h[j]) + sh[j];
+          if (f < 0.f) f = 0.f;
+          if (f > best[j]) {
+            best[j] = f;
+            bcode[j] = pc;
+          }
+        }
+      }
+    }
+    const long long ooff = row * C + (long long)lane * VEC;
+    BVec o;
+    CVec cv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      o.h[j] = __float2bfloat16(best[j]);
+      cv.c[j] = bcode[j];
+    }
+    *reinterpret_cast<uint4*>(y + ooff) = o.u;
+    *reinterpret_cast<uint2*>(code + ooff) = cv.u;
+  }
+}
+
 // Backward, one 2x2 INPUT tile per thread-lane. The tile at input
 // origin (2i, 2j) is touched only by the four output windows
 // (i,j), (i,j+1), (i+1,j), (i+1,j+1), and each of the tile's pixels sits
@@ -184,6 +262,20 @@ static int mp_grid(long long rows, int C) {
   long long blocks = (rows + rows_per_iter - 1) / rows_per_iter;
   if (blocks > 2080) blocks = 2080;
   return (int)(blocks > 0 ? blocks : 1);
+}
+
+extern "C" void launch_bn_maxpool_fwd(const void* x, const float* mean,
+                                      const float* invstd,
+                                      const float* weight,
+                                      const float* bias, void* y,
+                                      void* code, int N, int H, int W,
+                                      int Ho, int Wo, int C,
+                                      hipStream_t stream) {
+  hipLaunchKernelGGL(bn_maxpool_fwd_kernel,
+                     dim3(mp_grid((long long)N * Ho * Wo, C)), dim3(256),
+                     0, stream, (const bf16*)x, mean, invstd, weight,
+                     bias, (bf16*)y, (unsigned char*)code, N, H, W, Ho,
+                     Wo, C);
 }
 
 extern "C" void launch_maxpool_fwd(const void* x, void* y,

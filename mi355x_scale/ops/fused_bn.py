@@ -128,6 +128,76 @@ class _FusedBNFn(torch.autograd.Function):
                 None if fuse_acc else dbias, None, None, None, None, None)
 
 
+class _FusedBNPoolFn(torch.autograd.Function):
+    """Stem fusion: relu(bn(x)) + maxpool 3x3/s2/p1 with the normalized
+    map never materialized (the recompute-mask backward only needs x).
+    Saves a full 340 MB write + read per flagship step."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var,
+                momentum, eps, training):
+        C = x.shape[1]
+        dev = x.device
+        if training:
+            rows = x.numel() // C
+            mean = torch.empty(C, dtype=torch.float32, device=dev)
+            invstd = torch.empty(C, dtype=torch.float32, device=dev)
+            partial = _C.bn_fwd_reduce(x, C)
+            _C.bn_fwd_finalize(partial, mean, invstd, running_mean,
+                               running_var, momentum, eps, rows, C, True)
+        else:
+            mean = running_mean.to(torch.float32)
+            invstd = torch.rsqrt(running_var.to(torch.float32) + eps)
+        n, _, h, w = x.shape
+        ho, wo = (h + 1) // 2, (w + 1) // 2
+        y = torch.empty((n, C, ho, wo), dtype=x.dtype, device=dev,
+                        memory_format=torch.channels_last)
+        code = torch.empty(n * ho * wo * C, dtype=torch.uint8, device=dev)
+        _C.bn_maxpool3x3s2_fwd(x, mean, invstd, weight, bias, y, code)
+        ctx.save_for_backward(x, mean, invstd, weight, bias, code)
+        return y
+
+    @staticmethod
+    def backward(ctx, dp):
+        x, mean, invstd, weight, bias, code = ctx.saved_tensors
+        C = x.shape[1]
+        rows = x.numel() // C
+        dev = x.device
+        if not dp.is_contiguous(memory_format=torch.channels_last):
+            dp = dp.contiguous(memory_format=torch.channels_last)
+        # un-pool: scatter the pooled grad to the argmax positions
+        dz = torch.empty_like(x)
+        _C.maxpool3x3s2_bwd(dp, code, dz)
+        # then the recompute-mask BN backward (fused_bn.hip rm kernels)
+        k = torch.empty(3 * C, dtype=torch.float32, device=dev)
+
+        def _grad_view(p):
+            g = p.grad
+            if (g is not None and g.is_cuda and g.is_contiguous()
+                    and g.dtype == torch.float32 and g.numel() == C):
+                return g
+            return None
+
+        import os
+        wg, bg = _grad_view(weight), _grad_view(bias)
+        fuse_acc = (torch.cuda.is_current_stream_capturing()
+                    and wg is not None and bg is not None
+                    and os.environ.get("MI355X_BN_FUSED_ACC", "1") == "1")
+        if fuse_acc:
+            dweight, dbias = wg, bg
+        else:
+            dweight = torch.empty(C, dtype=torch.float32, device=dev)
+            dbias = torch.empty(C, dtype=torch.float32, device=dev)
+        dx = torch.empty_like(x)
+        partial = _C.bn_bwd_reduce_rm(dz, x, mean, invstd, weight, bias,
+                                      x.new_empty(0), C)
+        _C.bn_bwd_finalize(partial, invstd, weight, dweight, dbias, k,
+                           rows, C, fuse_acc)
+        _C.bn_bwd_apply_rm(dz, x, mean, invstd, bias, k, dx, C)
+        return (dx, None if fuse_acc else dweight,
+                None if fuse_acc else dbias, None, None, None, None, None)
+
+
 class FusedBNReLU2d(nn.Module):
     """BatchNorm2d fused with an optional residual add and ReLU.
 
@@ -157,6 +227,21 @@ class FusedBNReLU2d(nn.Module):
     def extra_repr(self) -> str:
         return (f"{self.num_features}, eps={self.eps}, "
                 f"momentum={self.momentum}, relu={self.relu}")
+
+    def forward_pooled(self, x: torch.Tensor) -> torch.Tensor:
+        """``maxpool3x3s2(relu(bn(x)))`` with the normalized map never
+        materialized (stem fusion; relu layers without residual only).
+        Falls back to the composed ops off the GPU bf16 path."""
+        assert self.relu, "pooled fusion is relu-only"
+        if _hip_supported(x):
+            require_ext()
+            if self.training:
+                self.num_batches_tracked += 1
+            x = x.contiguous(memory_format=torch.channels_last)
+            return _FusedBNPoolFn.apply(
+                x, self.weight, self.bias, self.running_mean,
+                self.running_var, self.momentum, self.eps, self.training)
+        return F.max_pool2d(self.forward(x), 3, stride=2, padding=1)
 
     def forward(self, x: torch.Tensor,
                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:

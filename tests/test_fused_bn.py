@@ -172,3 +172,55 @@ def test_gpu_resnet18_step_runs_fused():
     assert not FusedBNReLU2d.gpu_fallbacks, (
         f"BN layers silently fell back to MIOpen on GPU: "
         f"{FusedBNReLU2d.gpu_fallbacks[:8]}")
+
+
+@pytest.mark.gpu
+def test_bn_pool_fused_matches_composed():
+    """Stem fusion (forward_pooled): maxpool(relu(bn(x))) without
+    materializing the normalized map — fwd/bwd parity vs the fp32 torch
+    composition."""
+    import torch.nn.functional as F
+    from mi355x_scale.ops.fused_bn import FusedBNReLU2d
+
+    torch.manual_seed(0)
+    dev = "cuda"
+    n, c, h, w = 8, 64, 32, 32
+    xf = torch.randn(n, c, h, w, device=dev)
+
+    # fp32-math reference on the SAME bf16-rounded inputs/grads (a pure
+    # fp32 input would flip argmax codes at near-ties and scatter the
+    # pooled grad to different pixels than any bf16 pipeline can)
+    xb0 = xf.to(torch.bfloat16)
+    bn_ref = torch.nn.BatchNorm2d(c).to(dev)
+    xr = xb0.float().requires_grad_(True)
+    out_ref = F.max_pool2d(F.relu(bn_ref(xr)), 3, stride=2, padding=1)
+    g16 = torch.randn_like(out_ref).to(torch.bfloat16)
+    out_ref.backward(g16.float())
+
+    m = FusedBNReLU2d(c).to(dev)
+    xb = xb0.to(memory_format=torch.channels_last).requires_grad_(True)
+    out = m.forward_pooled(xb)
+    out.backward(g16.to(memory_format=torch.channels_last))
+    torch.cuda.synchronize()
+
+    rel = ((out.float() - out_ref).norm() / out_ref.norm()).item()
+    assert rel < 2e-2, f"fwd rel {rel}"
+    relx = ((xb.grad.float() - xr.grad).norm() / xr.grad.norm()).item()
+    assert relx < 3e-2, f"dx rel {relx}"
+    relw = ((m.weight.grad - bn_ref.weight.grad).norm()
+            / bn_ref.weight.grad.norm()).item()
+    assert relw < 3e-2, f"dweight rel {relw}"
+    relb = ((m.bias.grad - bn_ref.bias.grad).norm()
+            / bn_ref.bias.grad.norm()).item()
+    assert relb < 3e-2, f"dbias rel {relb}"
+    # running stats updated like the reference
+    assert torch.allclose(m.running_mean, bn_ref.running_mean, atol=2e-2)
+    assert torch.allclose(m.running_var, bn_ref.running_var, atol=2e-2)
+    # eval path uses running stats
+    m.eval()
+    bn_ref.eval()
+    with torch.no_grad():
+        oe = m.forward_pooled(xb.detach())
+        oer = F.max_pool2d(F.relu(bn_ref(xf)), 3, stride=2, padding=1)
+    rel_e = ((oe.float() - oer).norm() / oer.norm()).item()
+    assert rel_e < 2e-2, f"eval rel {rel_e}"
