@@ -277,7 +277,7 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
     int Nb, int H, int W, int HO, int WO, int rows_per_block,
     int phase_mask) {
   __shared__ bf16 xt[8 * XTROW];
-  __shared__ bf16 Dy[PXPAD * CO];
+  __shared__ bf16 Dy[2 * PXPAD * CO];  // double-buffered dy row slices
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -286,6 +286,10 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
   const int nrows = Nb * HO;
   const int mw = (wave >> 1) * 2;   // this wave's co-subtiles: mw, mw+1
   const int cw = (wave & 1) * 5;    // this wave's k-subtiles: cw .. cw+4
+  const int row_limit = min((blockIdx.x + 1) * rows_per_block, nrows);
+  const int nch = (WO * CO) >> 3;
+  const bf16x8 zv = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f,
+                     (bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
 
   f32x4 acc[2][5];
 #pragma unroll
@@ -293,38 +297,65 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
 #pragma unroll
     for (int c = 0; c < 5; ++c) acc[m][c] = (f32x4)0.0f;
 
+  // dy is SOFTWARE-PIPELINED: row r+1's slice is loaded into registers
+  // while row r's MFMA phase runs, and written to the other Dy buffer
+  // after it — the per-row global-load stall (the largest wrw phase in
+  // the ablation) hides under compute. Prologue: stage row 0 directly.
+  int cur = 0;
+  {
+    const int row0 = blockIdx.x * rows_per_block;
+    if (row0 < nrows) {
+      const bf16x8* src = reinterpret_cast<const bf16x8*>(
+          dy + (long long)row0 * WO * CO);
+      bf16x8* d8 = reinterpret_cast<bf16x8*>(Dy);
+      bf16x8 v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int i = tid + u * 256;
+        v[u] = (i < nch) ? src[i] : zv;
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) d8[tid + u * 256] = v[u];
+    }
+  }
+
   for (int rr = 0; rr < rows_per_block; ++rr) {
     const int row = blockIdx.x * rows_per_block + rr;
     if (row >= nrows) break;
     const int n = row / HO;
     const int ho = row - n * HO;
     const bool prime = (rr == 0) || (ho == 0);
-    __syncthreads();  // previous iteration's readers are done
+    __syncthreads();  // previous iteration's readers + Dy writes done
     if (prime)  // lead/tail pads must be zero before the first data rows
       for (int i = tid; i < 8 * XTROW; i += 256) xt[i] = (bf16)0.0f;
     __syncthreads();
     if (phase_mask & 1)
       stage_band_ring(x, xt, n, ho, H, W, tid, 256, prime);
-    if (phase_mask & 2) {
-      // dy row: [WO][CO] bf16 contiguous, 16B-aligned rows; 16B chunks
-      // register-batched 4 deep (4 x 256 x 8 = PXPAD*CO exactly)
+    // prefetch NEXT row's dy slice (latency overlaps band stage,
+    // barrier and the MFMA phase below)
+    bf16x8 v[4];
+    const bool have_next = (phase_mask & 2) && row + 1 < row_limit;
+    if (have_next) {
       const bf16x8* src = reinterpret_cast<const bf16x8*>(
-          dy + (long long)row * WO * CO);
-      bf16x8* d8 = reinterpret_cast<bf16x8*>(Dy);
-      const int nch = (WO * CO) >> 3;
-      const bf16x8 z = {(bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f,
-                        (bf16)0.0f, (bf16)0.0f, (bf16)0.0f, (bf16)0.0f};
-      bf16x8 v[4];
+          dy + (long long)(row + 1) * WO * CO);
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int i = tid + u * 256;
-        v[u] = (i < nch) ? src[i] : z;
+        v[u] = (i < nch) ? src[i] : zv;
       }
-#pragma unroll
-      for (int u = 0; u < 4; ++u) d8[tid + u * 256] = v[u];
     }
     __syncthreads();
-    if (!(phase_mask & 4)) continue;
+    if (phase_mask & 4) {
+      // (MFMA phase reads Dy[cur], written one iteration ago)
+    } else {
+      if (have_next) {
+        bf16x8* d8 = reinterpret_cast<bf16x8*>(Dy + (cur ^ 1) * PXPAD * CO);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) d8[tid + u * 256] = v[u];
+        cur ^= 1;
+      }
+      continue;
+    }
 
 #pragma unroll
     for (int sl = 0; sl < 4; ++sl) {
@@ -333,9 +364,10 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
 #pragma unroll
       for (int m = 0; m < 2; ++m) {
         bf16x8 f;
+        const bf16* dycur = Dy + cur * PXPAD * CO;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          f[j] = Dy[(pxbase + kgrp * 8 + j) * CO + (mw + m) * 16 + row16];
+          f[j] = dycur[(pxbase + kgrp * 8 + j) * CO + (mw + m) * 16 + row16];
         afrag[m] = f;
       }
 #pragma unroll
@@ -357,6 +389,12 @@ extern "C" __global__ __launch_bounds__(256) void stem_conv_wrw_kernel(
           acc[m][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[m], bfr, acc[m][c], 0, 0, 0);
       }
+    }
+    if (have_next) {
+      bf16x8* d8 = reinterpret_cast<bf16x8*>(Dy + (cur ^ 1) * PXPAD * CO);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) d8[tid + u * 256] = v[u];
+      cur ^= 1;
     }
   }
 
