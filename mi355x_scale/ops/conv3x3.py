@@ -23,7 +23,7 @@ NOTES_NEXT.md.
 
 ``Conv3x3`` subclasses ``nn.Conv2d`` (state-dict compatible, same
 init). When enabled, the HIP path takes CUDA bf16 channels-last with
-Cin == Cout ∈ {32..2048 step 32}, 9 <= W <= 64; everything else falls
+Cin == Cout ∈ {64..2048 step 64}, 9 <= W <= 64; everything else falls
 back to the standard conv autograd.
 """
 from __future__ import annotations

@@ -14,6 +14,12 @@ composition below doubles as the numerics reference used by the tests.
 State-dict layout matches ``nn.BatchNorm2d`` (weight, bias, running_mean,
 running_var, num_batches_tracked) so checkpoints interoperate.
 
+Round 2: the non-residual relu backward recomputes its mask as
+``(w*xhat + b) > 0`` from the saved x (``bn_bwd_*_rm`` kernels) — the
+stored y is read by NEITHER backward pass and is not saved; and the
+stem's ``forward_pooled`` folds the affine+relu into the maxpool window
+reads so the normalized stem map is never materialized at all.
+
 All device work is stream-ordered with no host sync, so the op captures
 into hipGraphs (train/graphstep.py).
 """

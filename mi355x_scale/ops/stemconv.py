@@ -1,10 +1,14 @@
 """MFMA stem convolution (7x7 stride-2 pad-3, 3->64, NHWC bf16).
 
 The ResNet stem is the one conv MIOpen leaves on a generic igemm tile
-(Cin=3 starves the GEMM K: K = 7*7*3 = 147) — measured ~298 us forward
-+ ~294 us weight-grad per step at bs 212 on the flagship bench, ~7% of
-the whole step. ``ops/csrc/stemconv.hip`` computes it as the GEMM it is
-on ``v_mfma_f32_16x16x32_bf16`` with wave-local LDS im2col tiles.
+(Cin=3 starves the GEMM K: K = 7*7*3 = 147) — MIOpen measured ~298 us
+forward + ~294 us weight-grad per step at bs 212 on the flagship bench,
+~7% of the whole step. ``ops/csrc/stemconv.hip`` computes it as the
+GEMM it is on ``v_mfma_f32_16x16x32_bf16`` with a pixel-padded LDS
+input band per output row: forward 210 us (1.73x MIOpen, padded
+weights read from L1-resident global), weight-grad 267 us (1.12x;
+ring-band staging + a software-pipelined dy stage prefetched into
+registers under the MFMA phase).
 
 ``StemConv2d`` subclasses ``nn.Conv2d`` (state-dict compatible, same
 init); the HIP path engages on CUDA bf16 channels-last inputs with the
