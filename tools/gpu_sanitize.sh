@@ -31,7 +31,8 @@ ASAN_RT=$(ldd "$SO" | awk '/libasan/ {print $3; exit}')
 echo "asan runtime: $ASAN_RT" | tee -a "$LOG"
 
 KTESTS="tests/test_batched_fit.py tests/test_fused_bn.py \
-tests/test_maxpool.py tests/test_flat_adam.py tests/test_preprocess.py"
+tests/test_maxpool.py tests/test_flat_adam.py tests/test_preprocess.py \
+tests/test_stemconv.py tests/test_conv3x3_wrw.py"
 
 export HSA_XNACK=1
 export MI355X_OPS_EXT=_C_asan
