@@ -45,10 +45,14 @@ def build_label_vocab(items: List[Tuple[str, str]]) -> Dict[str, int]:
             enumerate(sorted({lbl for _, lbl in items}))}
 
 
-def _decode_resize(path: str, hw: Tuple[int, int]) -> Optional[bytes]:
+def _decode_resize(path: str, hw: Tuple[int, int],
+                   store_format: str = "raw") -> Optional[bytes]:
     """PIL decode → resize(short side) → center-crop — the reference's
     CPU transform (``deep_learning/2...py:282-296``) applied ONCE at
-    ingest instead of on every epoch."""
+    ingest instead of on every epoch. ``store_format="jpeg"`` re-encodes
+    the cropped image (≈10x smaller rows; the reader pool then pays the
+    per-row decode the reference pays — feeds the ``image_format="jpeg"``
+    loader path and ``benchmarks/bench_loader_decode.py``)."""
     from PIL import Image
     h, w = hw
     try:
@@ -60,6 +64,11 @@ def _decode_resize(path: str, hw: Tuple[int, int]) -> Optional[bytes]:
             left = (nw - w) // 2
             top = (nh - h) // 2
             im = im.crop((left, top, left + w, top + h))
+            if store_format == "jpeg":
+                import io
+                buf = io.BytesIO()
+                im.save(buf, "JPEG", quality=90)
+                return buf.getvalue()
             return np.asarray(im, dtype=np.uint8).tobytes()
     except Exception:
         return None
@@ -72,10 +81,21 @@ def ingest_image_directory(
     rows_per_group: int = 212,
     rows_per_file: int = 2120,
     workers: int = 32,
+    store_format: str = "raw",
 ) -> Dict:
     """Decode every image under ``src_dir`` on a thread pool and write the
     parquet dataset + ``label_vocab.json``. Returns a manifest summary
-    (files, rows, skipped)."""
+    (files, rows, skipped).
+
+    ``store_format``: ``"raw"`` (default) stores fixed-size uint8 HWC
+    rows at training resolution (decode paid once — the streaming path
+    then reshapes bytes); ``"jpeg"`` stores re-encoded JPEG bytes
+    (variable binary; ≈10x smaller on disk, and training pays the
+    reference's per-row decode — pair with
+    ``ImageStreamDataModule(image_format="jpeg")``).
+    """
+    if store_format not in ("raw", "jpeg"):
+        raise ValueError("store_format must be 'raw' or 'jpeg'")
     os.makedirs(out_dir, exist_ok=True)
     items = discover_images(src_dir)
     if not items:
@@ -94,8 +114,10 @@ def ingest_image_directory(
         nonlocal file_idx, total
         if not buf_img:
             return
+        img_type = (pa.binary(row_bytes) if store_format == "raw"
+                    else pa.binary())
         table = pa.table({
-            "image": pa.array(buf_img, type=pa.binary(row_bytes)),
+            "image": pa.array(buf_img, type=img_type),
             "label": pa.array(buf_lbl, type=pa.int64()),
         })
         path = os.path.join(out_dir, f"part-{file_idx:05d}.parquet")
@@ -108,8 +130,10 @@ def ingest_image_directory(
 
     with ThreadPoolExecutor(max_workers=workers) as pool:
         for (path, label), blob in zip(
-                items, pool.map(lambda it: _decode_resize(it[0], image_hw),
-                                items, chunksize=16)):
+                items, pool.map(
+                    lambda it: _decode_resize(it[0], image_hw,
+                                              store_format),
+                    items, chunksize=16)):
             if blob is None:
                 skipped += 1
                 continue

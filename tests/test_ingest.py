@@ -53,3 +53,38 @@ def test_vocab_contiguous():
     items = [("a/x.jpg", "z"), ("b/y.jpg", "a"), ("c/z.jpg", "m")]
     v = build_label_vocab(items)
     assert v == {"a": 0, "m": 1, "z": 2}
+
+
+def test_ingest_jpeg_format_feeds_decode_path(tmp_path):
+    """store_format='jpeg': rows are encoded JPEG bytes consumable by
+    the image_format='jpeg' streaming loader."""
+    import numpy as np
+    import torch
+    from PIL import Image
+
+    from mi355x_scale.data.ingest import ingest_image_directory
+    from mi355x_scale.train import ImageStreamDataModule
+
+    src = tmp_path / "src"
+    rng = np.random.default_rng(0)
+    for cls in ("cat", "dog"):
+        (src / cls).mkdir(parents=True)
+        for i in range(6):
+            yy, xx = np.mgrid[0:80, 0:90]
+            img = np.clip(np.stack([127 + 100 * np.sin(xx / 9.0),
+                                    127 + 100 * np.cos(yy / 11.0),
+                                    rng.integers(0, 255, (80, 90))],
+                                   axis=-1), 0, 255).astype(np.uint8)
+            Image.fromarray(img).save(src / cls / f"{i}.jpg")
+    out = tmp_path / "ds"
+    summary = ingest_image_directory(str(src), str(out), image_hw=(64, 64),
+                                     rows_per_group=4, rows_per_file=8,
+                                     store_format="jpeg")
+    assert summary["rows"] == 12 and summary["skipped"] == 0
+    dm = ImageStreamDataModule(str(out), batch_size=4, workers_count=2,
+                               image_format="jpeg", image_hw=(64, 64),
+                               device=torch.device("cpu"))
+    batch = next(iter(dm.train_dataloader()))
+    assert batch["image"].shape == (4, 64, 64, 3)
+    assert batch["label"].max() <= 1
+    dm.teardown()
