@@ -139,7 +139,11 @@ def run_fine_grained_forecast_gpu(demand_df: pd.DataFrame,
     res = long_from_panel(y, gindex, dates, ["Product", "SKU"], "Date",
                           [("Demand", y.astype(np.float64)),
                            ("Demand_Fitted", fitted.astype(np.float64))])
-    res["Date"] = pd.to_datetime(res["Date"])
+    if not np.issubdtype(res["Date"].dtype, np.datetime64):
+        # tiled tvals are already datetime64 for datetime inputs —
+        # an unconditional to_datetime copied 15.7M timestamps (~0.11 s
+        # of the 0.72 s end-to-end job) for nothing
+        res["Date"] = pd.to_datetime(res["Date"])
     return res[["Product", "SKU", "Date", "Demand", "Demand_Fitted"]]
 
 
