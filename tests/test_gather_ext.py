@@ -136,3 +136,32 @@ try:
         assert (np.asarray(uniq) == np.asarray(ref_uniq)).all()
 except ImportError:  # pragma: no cover
     pass
+
+
+def test_panel_three_keys_matches_pandas_path():
+    """Guards the in-place combined-key build beyond two keys."""
+    rng = np.random.default_rng(3)
+    n = 4000
+    df = pd.DataFrame({
+        "A": rng.choice(["x", "y"], n),
+        "B": rng.choice([f"b{i}" for i in range(7)], n),
+        "C": rng.choice([f"c{i}" for i in range(5)], n),
+        "t": rng.integers(0, 10, n).astype(np.int64),
+        "v": rng.standard_normal(n).astype(np.float32),
+    }).drop_duplicates(["A", "B", "C", "t"])
+    from mi355x_scale.groupby.gather import panel_from_long
+    panel, gindex, tvals = panel_from_long(df, ["A", "B", "C"], "t", "v")
+    # oracle: pandas pivot on the MultiIndex
+    ref = df.set_index(["A", "B", "C", "t"])["v"]
+    for gi, key in enumerate(gindex):
+        for ti, t in enumerate(tvals):
+            want = ref.get(key + (t,))
+            got = panel[gi, ti]
+            if want is None:
+                assert np.isnan(got)
+            else:
+                assert got == np.float32(want)
+    # group index sorted like pandas MultiIndex factorize
+    codes_ref, idx_ref = pd.factorize(
+        pd.MultiIndex.from_frame(df[["A", "B", "C"]]), sort=True)
+    assert list(gindex) == list(idx_ref)
